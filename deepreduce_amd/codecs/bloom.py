@@ -1,0 +1,189 @@
+"""Bloom-filter index codec (GPU-first, device-side hashing).
+
+Reference behavior: /root/reference/pytorch/deepreduce.py:431-555 (GPU bloom
+driven by a precomputed 1 GB hash table + cupy packbits) and the C++ CPU op
+tensorflow/bloom_filter_compression.cc.  This build replaces the table with
+in-kernel MurmurHash3 double hashing (deepreduce_amd.hashing) and keeps the
+whole pipeline on-device.
+
+Filter config (matches pytorch/deepreduce.py:495-500):
+    h_f      = log2(1/fpr)            (float)
+    num_hash = ceil(h_f)
+    num_bits = ceil(h_f * capacity / ln 2)
+Default fpr = 0.1 * num_indices / grad_size (pytorch/deepreduce.py:511).
+
+Policies (pytorch/deepreduce.py:479-492 + tensorflow/policies.hpp):
+    'leftmost'      first K positives (ascending)
+    'random'        K positives drawn with a fixed-seed permutation
+    'p0'            ALL positives; true count travels prepended to vals
+    'conflict_sets' one element per hash-conflict set, round-robin (parity
+                    with policies.hpp:136-146; deterministic tie-breaks)
+
+Determinism contract: decompress re-derives indices on every rank from the
+bit array alone — policy + hashing must be bit-identical everywhere.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+
+from .. import ops
+from ..hashing import bloom_positions
+from . import SparseCompressor
+
+LN2 = 0.6931471805599453
+
+
+def get_bf_config(capacity: int, fpr: float):
+    """(num_hash, num_bits).  num_bits uses the UNceiled hash count, like the
+    reference (pytorch/deepreduce.py:495-500)."""
+    h_f = math.log2(1.0 / fpr)
+    num_hash = math.ceil(h_f)
+    num_bits = math.ceil(h_f * capacity / LN2)
+    return num_hash, max(num_bits, 8)
+
+
+def _policy_select(positives: torch.Tensor, k: int, policy: str, params, m: int, num_hash: int):
+    if policy == "leftmost":
+        return positives[:k]
+    if policy == "random":
+        seed = int(params.get("policy_seed", 42))
+        g = torch.Generator(device="cpu")
+        g.manual_seed(seed)
+        keys = torch.randperm(positives.numel(), generator=g)[:k].to(positives.device)
+        return positives[keys]
+    if policy == "p0":
+        return positives
+    if policy == "conflict_sets":
+        return _conflict_sets_select(positives, k, int(params.get("policy_seed", 42)), m, num_hash)
+    raise ValueError(f"unknown bloom policy {policy!r}")
+
+
+def _conflict_sets_select(positives: torch.Tensor, k: int, seed: int, m: int, num_hash: int):
+    """Deterministic conflict-sets policy (policies.hpp:43-146 semantics).
+
+    Positives are grouped by each of their hash values; sets ordered by
+    (size, smallest member); one element per set round-robin with a
+    minstd LCG.  Pure-python over the positive set only (small: ~K*(1+fpr)).
+    """
+    pos = positives.cpu().tolist()
+    hpos = bloom_positions(positives.cpu(), num_hash, m)  # [n, k]
+    sets: dict[int, set] = {}
+    for i, item in enumerate(pos):
+        for j in range(num_hash):
+            sets.setdefault(int(hpos[i, j]), set()).add(item)
+    ordered = sorted(sets.values(), key=lambda s: (len(s), min(s)))
+    ordered = [sorted(s) for s in ordered]
+    state = (seed % 2147483647) or 1
+
+    def lcg(n):
+        nonlocal state
+        state = (state * 48271) % 2147483647
+        return state % n
+
+    selected: set = set()
+    left = k
+    while left > 0:
+        progressed = False
+        for cset in ordered:
+            if left <= 0:
+                break
+            cset[:] = [x for x in cset if x not in selected]
+            if cset:
+                pick = cset.pop(lcg(len(cset)))
+                selected.add(pick)
+                left -= 1
+                progressed = True
+        if not progressed:
+            break
+    out = sorted(selected)
+    return torch.as_tensor(out, dtype=torch.int64, device=positives.device)
+
+
+class Bloom(SparseCompressor):
+    order_preserving = False
+
+    @staticmethod
+    def _config(num_indices: int, grad_size: int, params):
+        fpr = params.get("fpr", 0.1 * num_indices / grad_size)
+        return get_bf_config(num_indices, fpr)
+
+    @staticmethod
+    def compress(sparse_tensor, params):
+        vals, idxs, shape = sparse_tensor
+        grad_size = int(torch.Size(shape).numel())
+        num_indices = int(vals.numel())
+        policy = params.get("policy", "leftmost")
+
+        num_hash, m = Bloom._config(num_indices, grad_size, params)
+        packed = ops.bloom_insert(idxs, m, num_hash)
+
+        # FP-aware value re-read: query the filter the way decompress will,
+        # and send the dense tensor's values AT THOSE positions so false
+        # positives carry their true gradient value instead of garbage
+        # (pytorch/deepreduce.py:519-523).
+        dense = params.get("dense_tensor", None)
+        if dense is not None:
+            positives = ops.bloom_query_positives(packed, m, num_hash, grad_size)
+            new_idxs = _policy_select(positives, num_indices, policy, params, m, num_hash)
+            vals = dense.reshape(-1)[new_idxs]
+
+        if policy == "p0":
+            count = torch.as_tensor([num_indices], dtype=vals.dtype, device=vals.device)
+            vals = torch.cat([count, vals], dim=0)
+
+        return vals, packed, shape
+
+    @staticmethod
+    def decompress(bf_sparse_tensor, params):
+        vals, packed, shape = bf_sparse_tensor
+        policy = params.get("policy", "leftmost")
+        if policy == "p0":
+            count, vals = vals.split([1, vals.numel() - 1])
+            num_indices = int(count.item())
+        else:
+            num_indices = int(vals.numel())
+        grad_size = int(torch.Size(shape).numel())
+
+        num_hash, m = Bloom._config(num_indices, grad_size, params)
+        positives = ops.bloom_query_positives(packed, m, num_hash, grad_size)
+        idxs = _policy_select(positives, num_indices, policy, params, m, num_hash)
+        if policy == "p0":
+            # every positive gets a value; vals were either FP-aware-read at
+            # exactly these positions or truncated/padded to match
+            n = idxs.numel()
+            if vals.numel() < n:
+                vals = torch.nn.functional.pad(vals, (0, n - vals.numel()))
+            else:
+                vals = vals[:n]
+        else:
+            n = idxs.numel()
+            if n < vals.numel():
+                vals = vals[:n]
+            elif n > vals.numel():  # under-full positives (shouldn't happen)
+                idxs = idxs[: vals.numel()]
+        return vals, idxs, shape
+
+
+class BloomCPU(Bloom):
+    """CPU-pinned variant (reference parity for 'bloom_cpu',
+    pytorch/deepreduce.py:693-736 — there pybloomfilter-backed; here the same
+    deterministic torch codec executed on CPU)."""
+
+    @staticmethod
+    def compress(sparse_tensor, params):
+        vals, idxs, shape = sparse_tensor
+        device = vals.device
+        p = dict(params)
+        if p.get("dense_tensor") is not None:
+            p["dense_tensor"] = p["dense_tensor"].cpu()
+        v, b, s = Bloom.compress((vals.cpu(), idxs.cpu(), shape), p)
+        return v.to(device), b.to(device), s
+
+    @staticmethod
+    def decompress(bf_sparse_tensor, params):
+        vals, packed, shape = bf_sparse_tensor
+        device = vals.device
+        v, i, s = Bloom.decompress((vals.cpu(), packed.cpu(), shape), params)
+        return v.to(device), i.to(device), s

@@ -1,0 +1,68 @@
+"""Residual (error-feedback) memory — GRACE-`Memory` equivalent.
+
+Reference behavior (the TF copy carries the exact math the PyTorch side
+delegates to GRACE for): compensate t <- beta*residual + gamma*t, update
+residual <- compensated - decompressed(own payload)
+(/root/reference/tensorflow/deepreduce.py:31-52).
+
+The residual is the subtle invariant of the whole pipeline: the order is
+always compensate -> compress -> update (SURVEY.md sect. 7 "Overlap
+correctness").  `update` decompresses the rank's OWN payload, which for
+FP-aware bloom codecs is not simply scatter(vals, idxs).
+"""
+from __future__ import annotations
+
+import torch
+
+__all__ = ["Memory", "NoneMemory", "ResidualMemory", "memory_registry"]
+
+
+class Memory:
+    def compensate(self, tensor: torch.Tensor, name: str) -> torch.Tensor:
+        return tensor
+
+    def update(self, tensor, name, compressor, tensor_compressed, ctx):
+        pass
+
+    def state_dict(self):
+        return {}
+
+    def load_state_dict(self, state):
+        pass
+
+
+class NoneMemory(Memory):
+    pass
+
+
+class ResidualMemory(Memory):
+    def __init__(self, beta: float = 1.0, gamma: float = 1.0):
+        self.residuals: dict[str, torch.Tensor] = {}
+        self.beta = beta
+        self.gamma = gamma
+
+    def compensate(self, tensor, name):
+        r = self.residuals.get(name)
+        if r is None:
+            return tensor if self.gamma == 1.0 else self.gamma * tensor
+        return self.beta * r + self.gamma * tensor
+
+    def update(self, tensor, name, compressor, tensor_compressed, ctx):
+        decompressed = compressor.decompress(tensor_compressed, ctx)
+        self.residuals[name] = tensor - decompressed
+
+    # Checkpoint support (absent in the reference — residuals were lost on
+    # restart, tensorflow/deepreduce.py:39; we do better).
+    def state_dict(self):
+        return {"residuals": self.residuals, "beta": self.beta, "gamma": self.gamma}
+
+    def load_state_dict(self, state):
+        self.residuals = state["residuals"]
+        self.beta = state["beta"]
+        self.gamma = state["gamma"]
+
+
+memory_registry = {
+    "none": NoneMemory,
+    "residual": ResidualMemory,
+}

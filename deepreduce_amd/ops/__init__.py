@@ -1,0 +1,106 @@
+"""Op dispatch: HIP/CDNA4 kernels on GPU, torch reference on CPU.
+
+The HIP extension (`deepreduce_amd._hip_ops`, built in-tree from ops/src/ by
+setup.py / __graft_entry__.build) is the ONLY execution path on a GPU: if a
+tensor lives on a CUDA (ROCm) device and the extension failed to load, the op
+raises instead of silently falling back to eager PyTorch.  Set
+DEEPREDUCE_ALLOW_EAGER=1 to override (debug only).
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+
+from . import reference as _ref
+
+_hip = None
+_hip_err: str | None = None
+
+
+def _load_hip():
+    global _hip, _hip_err
+    if _hip is not None or _hip_err is not None:
+        return _hip
+    try:
+        from deepreduce_amd import _hip_ops  # built in-tree .so
+
+        _hip = _hip_ops
+    except Exception as e:  # noqa: BLE001
+        _hip_err = str(e)
+    return _hip
+
+
+def hip_available() -> bool:
+    return _load_hip() is not None
+
+
+def _want_hip(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    if _load_hip() is not None:
+        return True
+    if os.environ.get("DEEPREDUCE_ALLOW_EAGER") == "1":
+        return False
+    raise RuntimeError(
+        "deepreduce_amd: tensor is on a GPU but the HIP extension "
+        f"deepreduce_amd._hip_ops is not loadable ({_hip_err}). Build it with "
+        "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950), or "
+        "set DEEPREDUCE_ALLOW_EAGER=1 to force the eager fallback (slow)."
+    )
+
+
+# --------------------------------------------------------------------------
+# public ops
+# --------------------------------------------------------------------------
+
+def topk_select(flat: torch.Tensor, k: int):
+    # torch.topk is already a native (rocPRIM-backed) GPU path on ROCm;
+    # a fused HIP select kernel can replace it later if profiling says so.
+    return _ref.topk_select(flat, k)
+
+
+def bloom_insert(idxs: torch.Tensor, m: int, num_hash: int) -> torch.Tensor:
+    if _want_hip(idxs):
+        return _hip.bloom_insert(idxs, m, num_hash)
+    return _ref.bloom_insert(idxs, m, num_hash)
+
+
+def bloom_query_positives(packed: torch.Tensor, m: int, num_hash: int, universe: int):
+    if _want_hip(packed):
+        return _hip.bloom_query_positives(packed, m, num_hash, universe)
+    return _ref.bloom_query_positives(packed, m, num_hash, universe)
+
+
+def bloom_query_members(packed: torch.Tensor, m: int, num_hash: int, items: torch.Tensor):
+    if _want_hip(packed):
+        return _hip.bloom_query_members(packed, m, num_hash, items)
+    return _ref.bloom_query_members(packed, m, num_hash, items)
+
+
+def pack_ints(values: torch.Tensor, nbits: int) -> torch.Tensor:
+    if _want_hip(values):
+        return _hip.pack_ints(values, nbits)
+    return _ref.pack_ints(values, nbits)
+
+
+def unpack_ints(stream: torch.Tensor, n: int, nbits: int) -> torch.Tensor:
+    if _want_hip(stream):
+        return _hip.unpack_ints(stream, n, nbits)
+    return _ref.unpack_ints(stream, n, nbits)
+
+
+def qsgd_quantize(vals: torch.Tensor, quantum_num: int, bucket_size: int):
+    if _want_hip(vals):
+        return _hip.qsgd_quantize(vals, quantum_num, bucket_size)
+    return _ref.qsgd_quantize(vals, quantum_num, bucket_size)
+
+
+def qsgd_dequantize(levels: torch.Tensor, norms: torch.Tensor, quantum_num: int, bucket_size: int):
+    if _want_hip(levels):
+        return _hip.qsgd_dequantize(levels, norms, quantum_num, bucket_size)
+    return _ref.qsgd_dequantize(levels, norms, quantum_num, bucket_size)
+
+
+pack_bitarray = _ref.pack_bitarray
+unpack_bitarray = _ref.unpack_bitarray

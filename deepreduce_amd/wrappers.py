@@ -1,0 +1,198 @@
+"""DeepReduce wrapper layer — the core of the framework.
+
+Wraps any sparsifying Compressor and post-compresses its (values, indices)
+output with pluggable codecs.  Reference behavior:
+/root/reference/pytorch/deepreduce.py:51-302 and README.md:44-48.
+
+  ValueCompressor   post-compresses the VALUES array   ('deepreduce':'value')
+  IndexCompressor   post-compresses the INDICES array  ('deepreduce':'index')
+  DeepReduce        both, glued by an explicit `mapping` permutation
+                    ('deepreduce':'both')
+
+Small sparse tensors bypass compression entirely (numel <= 1000, matching
+pytorch/deepreduce.py:68).  `micro-benchmark` in params enables per-stage
+timing/volume prints (pytorch/deepreduce.py:74-95).
+"""
+from __future__ import annotations
+
+import time
+
+import torch
+
+from .codecs import compressor as codec_registry
+from .compressors import Compressor
+from .helper import tensor_bits
+
+_BYPASS_NUMEL = 1000
+
+
+def _sync_if(t: torch.Tensor):
+    if t.is_cuda:
+        torch.cuda.synchronize()
+
+
+class _WrapperBase(Compressor):
+    def __init__(self, sparsifier, params=None):
+        super().__init__(tensors_size_are_same=sparsifier.tensors_size_are_same)
+        self.average = getattr(sparsifier, "average", True)
+        self.sparsifier = sparsifier
+        self.params = params or {}
+
+    def aggregate(self, tensors):
+        return self.sparsifier.aggregate(tensors)
+
+
+class ValueCompressor(_WrapperBase):
+    """Sparsify, then post-compress the values array.
+
+    Parity: pytorch/deepreduce.py:51-97.  polyfit keeps uniform payload
+    sizes ragged-free only with the 'both'-style segment count — per-rank
+    num_pos changes segment counts, so like the reference (:364-367 note)
+    any value codec other than 'polyfit' marks payloads ragged; polyfit
+    itself is ragged too whenever ranks disagree on segment count, so we
+    conservatively mark ragged unless the sparsifier says otherwise AND the
+    codec is order/size stable ('qsgd').
+    """
+
+    def __init__(self, sparsifier, params=None):
+        super().__init__(sparsifier, params)
+        name = self.params.get("value", "polyfit")
+        self.val_codec = codec_registry[name]
+        if name not in ("qsgd",):
+            # polyfit coefficient count varies with per-rank num_pos
+            self.tensors_size_are_same = False
+
+    def compress(self, tensor, name):
+        tensors, ctx = self.sparsifier.compress(tensor, name)
+        vals, idxs = tensors
+        shape = ctx
+        if torch.Size(shape).numel() > _BYPASS_NUMEL:
+            start = time.perf_counter()
+            vals, idxs, shape_out = self.val_codec.compress((vals, idxs, tensor.size()), self.params)
+            if self.params.get("micro-benchmark", False):
+                _sync_if(tensor)
+                print(f"val_compression time:{time.perf_counter() - start}")
+            ctx = shape_out
+        return (vals, idxs), ctx
+
+    def decompress(self, tensors, ctx):
+        shape = ctx
+        vals, idxs = tensors
+        if torch.Size(shape).numel() > _BYPASS_NUMEL:
+            start = time.perf_counter()
+            vals, idxs, shape = self.val_codec.decompress((vals, idxs, shape), self.params)
+            if self.params.get("micro-benchmark", False):
+                print(f"val_decompression time:{time.perf_counter() - start}")
+                dense_bits = torch.Size(shape).numel() * 32
+                print(f"idx_relative_volume: {tensor_bits([tensors[1]]) / dense_bits:.4f}")
+                print(f"val_relative_volume: {tensor_bits([tensors[0]]) / dense_bits:.4f}")
+        return self.sparsifier.decompress((vals, idxs), shape)
+
+
+class IndexCompressor(_WrapperBase):
+    """Sparsify, then post-compress the indices array.
+
+    Parity: pytorch/deepreduce.py:100-153.  The dense tensor rides along in
+    params for the Bloom codec's FP-aware value re-read (:117).
+    """
+
+    def __init__(self, sparsifier, params=None):
+        super().__init__(sparsifier, params)
+        name = self.params.get("index", "bloom")
+        self.idx_codec = codec_registry[name]
+        if name not in ("bloom",) or self.params.get("policy") == "p0":
+            self.tensors_size_are_same = False
+
+    def compress(self, tensor, name):
+        tensors, ctx = self.sparsifier.compress(tensor, name)
+        vals, idxs = tensors
+        shape = ctx
+        if torch.Size(shape).numel() > _BYPASS_NUMEL:
+            self.params["dense_tensor"] = tensor
+            start = time.perf_counter()
+            vals, idxs, shape_out = self.idx_codec.compress((vals, idxs, tensor.size()), self.params)
+            self.params.pop("dense_tensor", None)
+            if self.params.get("micro-benchmark", False):
+                _sync_if(tensor)
+                print(f"idx_compression time:{time.perf_counter() - start}")
+            ctx = shape_out
+        return (vals, idxs), ctx
+
+    def decompress(self, tensors, ctx):
+        shape = ctx
+        vals, idxs = tensors
+        if torch.Size(shape).numel() > _BYPASS_NUMEL:
+            start = time.perf_counter()
+            vals, idxs, shape = self.idx_codec.decompress((vals, idxs, shape), self.params)
+            if self.params.get("micro-benchmark", False):
+                print(f"idx_decompression time:{time.perf_counter() - start}")
+                dense_bits = torch.Size(shape).numel() * 32
+                print(f"idx_relative_volume: {tensor_bits([tensors[1]]) / dense_bits:.4f}")
+                print(f"val_relative_volume: {tensor_bits([tensors[0]]) / dense_bits:.4f}")
+        return self.sparsifier.decompress((vals, idxs), shape)
+
+
+class DeepReduce(_WrapperBase):
+    """Joint index+value compression glued by a mapping permutation.
+
+    Parity: pytorch/deepreduce.py:156-302.  compress: index codec first
+    (bloom bits), then the value codec over the Bloom-ordered values with
+    new_idxs = arange — its non-order-preserving sort permutation becomes
+    `mapping`.  decompress: values from coeffs (in sorted order), indices
+    re-derived from the bloom bits (ascending), then idxs = idxs[mapping]
+    undoes the value sort (:290).  Wire = (vals', bloom_bits, mapping).
+    """
+
+    def __init__(self, sparsifier, params=None):
+        super().__init__(sparsifier, params)
+        self.val_codec = codec_registry[self.params.get("value", "polyfit")]
+        self.idx_codec = codec_registry[self.params.get("index", "bloom")]
+        self.tensors_size_are_same = False  # coeff counts vary with num_pos
+
+    def compress(self, tensor, name):
+        tensors, ctx = self.sparsifier.compress(tensor, name)
+        vals, idxs = tensors
+        shape = ctx
+        start = time.perf_counter()
+        if torch.Size(shape).numel() > _BYPASS_NUMEL:
+            # FP-aware re-read (improvement over the reference, which skips it
+            # in 'both' mode): values are read from the dense tensor at the
+            # positions decompress will deterministically re-derive, so false
+            # positives no longer shift the index<->value alignment.
+            if self.params.get("fp_aware", True):
+                self.params["dense_tensor"] = tensor
+            vals, idxs, _ = self.idx_codec.compress((vals, idxs, tensor.size()), self.params)
+            self.params.pop("dense_tensor", None)
+            new_idxs = torch.arange(vals.numel(), device=vals.device)
+            vals, mapping, shape_out = self.val_codec.compress((vals, new_idxs, shape), self.params)
+            ctx = shape_out
+            tensors = (vals, idxs, mapping.int())
+        if self.params.get("micro-benchmark", False):
+            _sync_if(tensor)
+            print(f"_compression time:{time.perf_counter() - start}")
+        return tensors, ctx
+
+    def decompress(self, tensors, ctx):
+        shape = ctx
+        start = time.perf_counter()
+        if torch.Size(shape).numel() > _BYPASS_NUMEL:
+            vals, idxs, mapping = tensors
+            mapping = mapping.long()
+            vals, _, _ = self.val_codec.decompress((vals, mapping, shape), self.params)
+            _, idxs, _ = self.idx_codec.decompress((mapping, idxs, shape), self.params)
+            idxs = idxs[mapping]
+        else:
+            vals, idxs = tensors
+        if self.params.get("micro-benchmark", False):
+            print(f"_decompression time:{time.perf_counter() - start}")
+            dense_bits = torch.Size(shape).numel() * 32
+            print(f"idx_relative_volume: {tensor_bits(list(tensors[1:])) / dense_bits:.4f}")
+            print(f"val_relative_volume: {tensor_bits([tensors[0]]) / dense_bits:.4f}")
+        return self.sparsifier.decompress((vals, idxs), shape)
+
+
+deepreduce_wrapper = {
+    "value": ValueCompressor,
+    "index": IndexCompressor,
+    "both": DeepReduce,
+}

@@ -1,0 +1,121 @@
+"""Bloom codec: no-false-negative property, FPR vs theory, policy
+determinism across 'ranks', FP-aware value re-read."""
+import math
+
+import pytest
+import torch
+
+from deepreduce_amd.codecs import compressor
+from deepreduce_amd.codecs.bloom import get_bf_config
+from deepreduce_amd.ops import (
+    bloom_insert,
+    bloom_query_members,
+    bloom_query_positives,
+    topk_select,
+)
+
+
+@pytest.fixture
+def data():
+    torch.manual_seed(3)
+    d = 100_000
+    t = torch.randn(d)
+    vals, idxs = topk_select(t, 1000)
+    return t, vals, idxs
+
+
+def test_no_false_negatives(data):
+    t, vals, idxs = data
+    nh, m = get_bf_config(1000, 0.001)
+    packed = bloom_insert(idxs, m, nh)
+    assert bloom_query_members(packed, m, nh, idxs).all()
+
+
+def test_fpr_close_to_theory(data):
+    t, vals, idxs = data
+    d = t.numel()
+    fpr = 0.001
+    nh, m = get_bf_config(1000, fpr)
+    packed = bloom_insert(idxs, m, nh)
+    pos = bloom_query_positives(packed, m, nh, d)
+    true = set(idxs.tolist())
+    fp = len(set(pos.tolist()) - true)
+    measured = fp / (d - len(true))
+    assert measured < fpr * 3  # generous: binomial noise
+    assert set(pos.tolist()) >= true  # superset (no false negatives)
+
+
+def test_positives_sorted_ascending(data):
+    t, vals, idxs = data
+    nh, m = get_bf_config(1000, 0.01)
+    packed = bloom_insert(idxs, m, nh)
+    pos = bloom_query_positives(packed, m, nh, t.numel())
+    assert torch.equal(pos, pos.sort().values)
+
+
+def test_config_matches_reference_formula():
+    # pytorch/deepreduce.py:495-500
+    nh, m = get_bf_config(1000, 0.001)
+    assert nh == math.ceil(math.log2(1000))
+    assert m == math.ceil(math.log2(1000) * 1000 / 0.693147180)
+
+
+@pytest.mark.parametrize("policy", ["leftmost", "random", "p0", "conflict_sets"])
+def test_compress_decompress_determinism(data, policy):
+    """decompress must re-derive identical indices on every 'rank'."""
+    t, vals, idxs = data
+    params = {"policy": policy, "dense_tensor": t}
+    v, bits, shape = compressor["bloom"].compress((vals, idxs, t.size()), params)
+    params2 = {"policy": policy}  # decompress side has no dense tensor
+    out1 = compressor["bloom"].decompress((v.clone(), bits.clone(), shape), params2)
+    out2 = compressor["bloom"].decompress((v.clone(), bits.clone(), shape), params2)
+    assert torch.equal(out1[1], out2[1])
+    assert torch.equal(out1[0], out2[0])
+
+
+@pytest.mark.parametrize("policy", ["leftmost", "p0"])
+def test_fp_aware_values_exact(data, policy):
+    """with FP-aware re-read, every (val, idx) pair matches the dense tensor."""
+    t, vals, idxs = data
+    params = {"policy": policy, "dense_tensor": t}
+    v, bits, shape = compressor["bloom"].compress((vals, idxs, t.size()), params)
+    v2, i2, _ = compressor["bloom"].decompress((v, bits, shape), {"policy": policy})
+    assert torch.allclose(t[i2], v2)
+
+
+def test_recall_with_fp_aware(data):
+    t, vals, idxs = data
+    params = {"policy": "leftmost", "dense_tensor": t, "fpr": 0.001}
+    v, bits, shape = compressor["bloom"].compress((vals, idxs, t.size()), params)
+    v2, i2, _ = compressor["bloom"].decompress((v, bits, shape), {"policy": "leftmost", "fpr": 0.001})
+    true = set(idxs.tolist())
+    rec = set(i2.tolist())
+    # leftmost drops ~#false-positives of the rightmost true indices:
+    # expected recall ~ 1 - fpr*d/k = 0.9 here (the paper's motivation for P0)
+    assert len(true & rec) / len(true) > 0.85
+
+
+def test_p0_returns_all_positives(data):
+    t, vals, idxs = data
+    params = {"policy": "p0", "dense_tensor": t}
+    v, bits, shape = compressor["bloom"].compress((vals, idxs, t.size()), params)
+    assert v.numel() >= idxs.numel() + 1  # count + all positives
+    v2, i2, _ = compressor["bloom"].decompress((v, bits, shape), {"policy": "p0"})
+    assert set(i2.tolist()) >= set(idxs.tolist())
+    assert v2.numel() == i2.numel()
+
+
+def test_conflict_sets_covers_k(data):
+    t, vals, idxs = data
+    params = {"policy": "conflict_sets", "dense_tensor": t}
+    v, bits, shape = compressor["bloom"].compress((vals, idxs, t.size()), params)
+    v2, i2, _ = compressor["bloom"].decompress((v, bits, shape), {"policy": "conflict_sets"})
+    assert i2.numel() == idxs.numel()
+    assert torch.allclose(t[i2], v2)
+
+
+def test_wire_volume_beats_raw_indices(data):
+    t, vals, idxs = data
+    params = {"policy": "leftmost", "fpr": 0.01}
+    v, bits, shape = compressor["bloom"].compress((vals, idxs, t.size()), params)
+    assert bits.numel() < idxs.numel() * 4  # paper: ~50% of int32 keys

@@ -1,0 +1,167 @@
+"""Multi-process plumbing tests on gloo CPU, world_size=2 — BASELINE.json
+config 1 (ResNet-20/CIFAR-10 top-k 1% + residual over allgather) plus the
+ragged-payload path.  No GPU required."""
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from deepreduce_amd.models import resnet20
+
+
+def _init(rank, world):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ.setdefault("MASTER_PORT", "29611")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+
+def _run_allgather_step(rank, world, mode, q):
+    try:
+        _init(rank, world)
+        from deepreduce_amd import deepreduce_from_params
+
+        params = {
+            "compressor": "topk",
+            "memory": "residual",
+            "communicator": "allgather",
+            "compress_ratio": 0.01,
+        }
+        if mode == "index":
+            params.update({"deepreduce": "index", "index": "bloom", "policy": "leftmost"})
+        elif mode == "both":
+            params.update({"deepreduce": "both", "index": "bloom", "value": "polyfit"})
+        elif mode == "value":
+            params.update({"deepreduce": "value", "value": "qsgd"})
+        grc = deepreduce_from_params(params)
+
+        torch.manual_seed(100 + rank)  # DIFFERENT grads per rank
+        g = torch.randn(8000)
+        out = grc.step(g, "w")
+        # every rank must compute the identical averaged tensor
+        outs = [torch.empty_like(out) for _ in range(world)]
+        dist.all_gather(outs, out)
+        same = all(torch.allclose(outs[0], o, atol=1e-5) for o in outs)
+        q.put((rank, bool(same), float(out.abs().sum())))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e!r}", 0.0))
+
+
+@pytest.mark.parametrize("mode", ["plain", "index", "value", "both"])
+def test_allgather_identical_average_across_ranks(mode):
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    os.environ["MASTER_PORT"] = str(29620 + hash(mode) % 50)
+    procs = [ctx.Process(target=_run_allgather_step, args=(r, world, mode, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, same, s in results:
+        assert same is True, f"rank {rank}: {same}"
+        assert s > 0
+
+
+def _run_resnet20_training(rank, world, q):
+    try:
+        _init(rank, world)
+        from deepreduce_amd import (
+            DistributedOptimizer,
+            broadcast_parameters,
+            deepreduce_from_params,
+        )
+
+        torch.manual_seed(0)  # same init everywhere, then broadcast anyway
+        model = resnet20()
+        broadcast_parameters(model)
+        params = {
+            "compressor": "topk",
+            "memory": "residual",
+            "communicator": "allgather",
+            "compress_ratio": 0.01,
+            "deepreduce": "index",
+            "index": "bloom",
+        }
+        grc = deepreduce_from_params(params)
+        opt = DistributedOptimizer(torch.optim.SGD(model.parameters(), lr=0.05), grc, model)
+
+        torch.manual_seed(1000 + rank)  # different data per rank
+        losses = []
+        for _ in range(3):
+            x = torch.randn(16, 3, 32, 32)
+            y = torch.randint(0, 10, (16,))
+            opt.zero_grad()
+            loss = torch.nn.functional.cross_entropy(model(x), y)
+            loss.backward()
+            opt.step()
+            losses.append(loss.item())
+        # parameters must stay in sync across ranks
+        p0 = next(model.parameters()).data.reshape(-1)[:100]
+        ps = [torch.empty_like(p0) for _ in range(world)]
+        dist.all_gather(ps, p0)
+        in_sync = all(torch.allclose(ps[0], p, atol=1e-5) for p in ps)
+        q.put((rank, bool(in_sync), losses, opt.last_wire_bytes))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e!r}", [], 0))
+
+
+def test_resnet20_cifar_topk_training_stays_in_sync():
+    """BASELINE.json config 1."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    os.environ["MASTER_PORT"] = "29701"
+    procs = [ctx.Process(target=_run_resnet20_training, args=(r, world, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=600) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, in_sync, losses, wire in results:
+        assert in_sync is True, f"rank {rank}: {in_sync}"
+        assert len(losses) == 3
+        assert wire > 0
+
+
+def _run_ragged(rank, world, q):
+    try:
+        _init(rank, world)
+        from deepreduce_amd import deepreduce_from_params
+
+        params = {
+            "compressor": "threshold",
+            "threshold": 0.5 + 0.3 * rank,  # DIFFERENT payload sizes per rank
+            "memory": "none",
+            "communicator": "allgather",
+        }
+        grc = deepreduce_from_params(params)
+        torch.manual_seed(55 + rank)
+        g = torch.randn(5000)
+        out = grc.step(g, "w")
+        outs = [torch.empty_like(out) for _ in range(world)]
+        dist.all_gather(outs, out)
+        same = all(torch.allclose(outs[0], o, atol=1e-5) for o in outs)
+        q.put((rank, bool(same)))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+def test_ragged_allgather_threshold():
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    os.environ["MASTER_PORT"] = "29702"
+    procs = [ctx.Process(target=_run_ragged, args=(r, world, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, same in results:
+        assert same is True, f"rank {rank}: {same}"

@@ -1,0 +1,147 @@
+"""Host runtime: sparsifiers, residual memory, factory/params contract,
+packing helpers, hashing invariants."""
+import torch
+
+from deepreduce_amd import (
+    DeepReduce,
+    IndexCompressor,
+    ResidualMemory,
+    TopKCompressor,
+    ValueCompressor,
+    deepreduce_from_params,
+    grace_from_params,
+    tensor_bits,
+)
+from deepreduce_amd.communicator import Allgather, _flatten_payload, _unflatten_payload
+from deepreduce_amd.compressors import ThresholdCompressor, RandomKCompressor
+from deepreduce_amd.hashing import fmix32
+
+
+def test_topk_sparsifier_roundtrip():
+    t = torch.randn(4, 100)
+    c = TopKCompressor(0.05)
+    (vals, idxs), ctx = c.compress(t, "w")
+    assert vals.numel() == 20
+    dense = c.decompress((vals, idxs), ctx)
+    assert dense.shape == t.shape
+    # kept entries exact, others zero
+    mask = torch.zeros(400, dtype=torch.bool)
+    mask[idxs] = True
+    assert torch.equal(dense.reshape(-1)[mask], t.reshape(-1)[mask])
+    assert (dense.reshape(-1)[~mask] == 0).all()
+
+
+def test_threshold_sparsifier():
+    t = torch.randn(1000)
+    c = ThresholdCompressor(0.5)
+    (vals, idxs), ctx = c.compress(t, "w")
+    assert (vals.abs() >= 0.5).all()
+    assert not c.tensors_size_are_same
+
+
+def test_randomk_deterministic_across_ranks():
+    t1, t2 = torch.randn(1000), torch.randn(1000)
+    c1, c2 = RandomKCompressor(0.1), RandomKCompressor(0.1)
+    (_, i1), _ = c1.compress(t1, "w")
+    (_, i2), _ = c2.compress(t2, "w")
+    assert torch.equal(i1, i2)  # same name+step -> same positions
+
+
+def test_residual_memory_error_feedback():
+    mem = ResidualMemory()
+    c = TopKCompressor(0.1)
+    t = torch.randn(500)
+    comp = mem.compensate(t, "w")
+    assert torch.equal(comp, t)  # first step: no residual
+    payload, ctx = c.compress(comp, "w")
+    mem.update(comp, "w", c, payload, ctx)
+    # residual + decompressed == compensated (exact for topk)
+    assert torch.allclose(mem.residuals["w"] + c.decompress(payload, ctx), comp)
+    # second step: compensation adds the residual
+    t2 = torch.randn(500)
+    comp2 = mem.compensate(t2, "w")
+    assert torch.allclose(comp2, mem.residuals["w"] + t2)
+
+
+def test_memory_checkpoint_roundtrip():
+    mem = ResidualMemory()
+    mem.residuals["w"] = torch.randn(10)
+    sd = mem.state_dict()
+    mem2 = ResidualMemory()
+    mem2.load_state_dict(sd)
+    assert torch.equal(mem2.residuals["w"], mem.residuals["w"])
+
+
+def test_grace_from_params_readme_contract():
+    """The README.md:37 params dict must work unchanged."""
+    params = {
+        "compressor": "topk",
+        "memory": "residual",
+        "communicator": "allgather",
+        "compress_ratio": 0.01,
+        "deepreduce": "index",
+        "index": "bloom",
+    }
+    grc = grace_from_params(params)
+    assert isinstance(grc, Allgather)
+    assert isinstance(grc.compressor, TopKCompressor)
+    # wrap-after-build, exactly as README.md:44-48 does
+    from deepreduce_amd.wrappers import deepreduce_wrapper
+
+    grc.compressor = deepreduce_wrapper[params["deepreduce"]](grc.compressor, params)
+    assert isinstance(grc.compressor, IndexCompressor)
+
+
+def test_deepreduce_from_params_modes():
+    base = {"compressor": "topk", "memory": "none", "communicator": "allgather",
+            "compress_ratio": 0.02}
+    assert isinstance(deepreduce_from_params({**base, "deepreduce": "value"}).compressor, ValueCompressor)
+    assert isinstance(deepreduce_from_params({**base, "deepreduce": "index"}).compressor, IndexCompressor)
+    assert isinstance(deepreduce_from_params({**base, "deepreduce": "both"}).compressor, DeepReduce)
+    assert isinstance(deepreduce_from_params(base).compressor, TopKCompressor)
+    # hash_table key accepted (and ignored) for reference compatibility
+    deepreduce_from_params({**base, "deepreduce": "index", "hash_table": None})
+
+
+def test_single_rank_step_end_to_end():
+    params = {"compressor": "topk", "memory": "residual", "communicator": "allgather",
+              "compress_ratio": 0.05, "deepreduce": "index", "index": "bloom"}
+    grc = deepreduce_from_params(params)
+    g = torch.randn(5000)
+    out = grc.step(g, "layer.weight")
+    assert out.shape == g.shape
+    assert (out != 0).sum() > 0
+
+
+def test_tensor_bits():
+    assert tensor_bits([torch.zeros(10, dtype=torch.float32)]) == 320
+    assert tensor_bits([torch.zeros(10, dtype=torch.uint8)]) == 80
+    assert tensor_bits([torch.zeros(3, dtype=torch.int64), torch.zeros(2, dtype=torch.int8)]) == 208
+
+
+def test_payload_flatten_roundtrip():
+    payload = (torch.randn(7), torch.arange(5, dtype=torch.int64),
+               torch.zeros(3, dtype=torch.uint8))
+    buf, metas = _flatten_payload(payload)
+    assert buf.dtype == torch.uint8
+    out = _unflatten_payload(buf, metas)
+    for a, b in zip(payload, out):
+        assert torch.equal(a, b)
+
+
+def test_fmix32_known_values():
+    # murmur3 fmix32 reference vectors
+    import numpy as np
+
+    def ref(h):
+        h ^= h >> 16
+        h = (h * 0x85EBCA6B) & 0xFFFFFFFF
+        h ^= h >> 13
+        h = (h * 0xC2B2AE35) & 0xFFFFFFFF
+        h ^= h >> 16
+        return h
+
+    xs = torch.tensor([0, 1, 2, 12345, 0xDEADBEEF, 0xFFFFFFFF], dtype=torch.int64)
+    out = fmix32(xs)
+    for x, o in zip(xs.tolist(), out.tolist()):
+        assert o == ref(x)

@@ -37,6 +37,9 @@ def get_args():
     p.add_argument("--index", default="bloom")
     p.add_argument("--policy", default="leftmost")
     p.add_argument("--device", default=None)
+    p.add_argument("--overlap", action="store_true",
+                   help="hook-driven compression overlapped with backward + "
+                        "fused dense exchange of small tensors")
     return p.parse_args()
 
 
@@ -104,9 +107,29 @@ def main():
     broadcast_parameters(model)
 
     grc, params = build_grc(args)
-    opt = DistributedOptimizer(
-        torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4), grc, model
-    )
+    sgd = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
+    reducer = None
+    if args.overlap and args.deepreduce != "dense":
+        from deepreduce_amd.parallel import OverlappedReducer
+
+        reducer = OverlappedReducer(model, grc)
+        opt = sgd  # reducer handles the exchange; plain SGD steps
+
+        class _Opt:
+            last_wire_bytes = 0
+
+            def zero_grad(self, set_to_none=False):
+                sgd.zero_grad(set_to_none=set_to_none)
+                reducer.zero_wire_counter()
+
+            def step(self):
+                reducer.finalize()
+                self.last_wire_bytes = reducer.last_wire_bytes
+                sgd.step()
+
+        opt = _Opt()
+    else:
+        opt = DistributedOptimizer(sgd, grc, model)
     loss_fn = (
         torch.nn.BCEWithLogitsLoss() if args.model == "ncf" else torch.nn.CrossEntropyLoss()
     )

@@ -128,6 +128,9 @@ class Bloom(SparseCompressor):
             positives = ops.bloom_query_positives(packed, m, num_hash, grad_size)
             new_idxs = _policy_select(positives, num_indices, policy, params, m, num_hash)
             vals = dense.reshape(-1)[new_idxs]
+            # side-channel for the wrappers' own-payload cache: decompress of
+            # this payload deterministically yields exactly (vals, new_idxs)
+            params["_own_decoded"] = (vals, new_idxs)
 
         if policy == "p0":
             count = torch.as_tensor([num_indices], dtype=vals.dtype, device=vals.device)

@@ -48,7 +48,14 @@ class ResidualMemory(Memory):
         return self.beta * r + self.gamma * tensor
 
     def update(self, tensor, name, compressor, tensor_compressed, ctx):
-        decompressed = compressor.decompress(tensor_compressed, ctx)
+        # Wrappers cache their own-payload decompression at compress time
+        # (decompress_own) so e.g. the Bloom full-universe query is not run
+        # twice per tensor per step; falls back to a full decompress.
+        own = getattr(compressor, "decompress_own", None)
+        if own is not None:
+            decompressed = own(tensor_compressed, ctx, name)
+        else:
+            decompressed = compressor.decompress(tensor_compressed, ctx)
         self.residuals[name] = tensor - decompressed
 
     # Checkpoint support (absent in the reference — residuals were lost on

@@ -37,9 +37,21 @@ class _WrapperBase(Compressor):
         self.average = getattr(sparsifier, "average", True)
         self.sparsifier = sparsifier
         self.params = params or {}
+        # single-slot cache: (name -> (vals, idxs, shape)) of THIS rank's
+        # payload in decoded form, filled during compress so the residual
+        # update (memory.update -> decompress_own) skips re-decoding —
+        # for bloom that saves one full-universe query per tensor per step.
+        self._own_cache: dict = {}
 
     def aggregate(self, tensors):
         return self.sparsifier.aggregate(tensors)
+
+    def decompress_own(self, tensors, ctx, name):
+        cached = self._own_cache.pop(name, None)
+        if cached is not None:
+            vals, idxs, shape = cached
+            return self.sparsifier.decompress((vals, idxs), shape)
+        return self.decompress(tensors, ctx)
 
 
 class ValueCompressor(_WrapperBase):
@@ -112,6 +124,9 @@ class IndexCompressor(_WrapperBase):
             start = time.perf_counter()
             vals, idxs, shape_out = self.idx_codec.compress((vals, idxs, tensor.size()), self.params)
             self.params.pop("dense_tensor", None)
+            own = self.params.pop("_own_decoded", None)
+            if own is not None:
+                self._own_cache[name] = (own[0], own[1], shape)
             if self.params.get("micro-benchmark", False):
                 _sync_if(tensor)
                 print(f"idx_compression time:{time.perf_counter() - start}")
@@ -163,6 +178,9 @@ class DeepReduce(_WrapperBase):
                 self.params["dense_tensor"] = tensor
             vals, idxs, _ = self.idx_codec.compress((vals, idxs, tensor.size()), self.params)
             self.params.pop("dense_tensor", None)
+            own = self.params.pop("_own_decoded", None)
+            if own is not None:
+                self._own_cache[name] = own[1]  # bloom-recovered indices
             new_idxs = torch.arange(vals.numel(), device=vals.device)
             vals, mapping, shape_out = self.val_codec.compress((vals, new_idxs, shape), self.params)
             ctx = shape_out
@@ -188,6 +206,19 @@ class DeepReduce(_WrapperBase):
             dense_bits = torch.Size(shape).numel() * 32
             print(f"idx_relative_volume: {tensor_bits(list(tensors[1:])) / dense_bits:.4f}")
             print(f"val_relative_volume: {tensor_bits([tensors[0]]) / dense_bits:.4f}")
+        return self.sparsifier.decompress((vals, idxs), shape)
+
+    def decompress_own(self, tensors, ctx, name):
+        """Own-payload decode with the cached bloom indices: only the (cheap)
+        value-codec eval runs; the full-universe query is skipped."""
+        cached_idxs = self._own_cache.pop(name, None)
+        shape = ctx
+        if cached_idxs is None or torch.Size(shape).numel() <= _BYPASS_NUMEL:
+            return self.decompress(tensors, ctx)
+        vals, idxs, mapping = tensors
+        mapping = mapping.long()
+        vals, _, _ = self.val_codec.decompress((vals, mapping, shape), self.params)
+        idxs = cached_idxs[mapping]
         return self.sparsifier.decompress((vals, idxs), shape)
 
 

@@ -61,7 +61,9 @@ def test_bloom_members_parity(hip, dev):
     gpu = hip.bloom_query_members(packed, m, k, probe)
     cpu = ref.bloom_query_members(packed.cpu(), m, k, probe.cpu())
     assert torch.equal(gpu.cpu(), cpu)
-    assert gpu[probe % 7 == 0].all()  # no false negatives
+    # no false negatives among probes that were actually inserted
+    inserted = (probe % 7 == 0) & (probe < 50_000)
+    assert gpu[inserted].all()
 
 
 def test_pack_unpack_parity(hip, dev):

@@ -100,3 +100,32 @@ def test_micro_benchmark_prints(grad, capsys):
     out = capsys.readouterr().out
     assert "val_compression time" in out
     assert "val_relative_volume" in out
+
+
+def test_decompress_own_matches_decompress(grad):
+    """the residual-update fast path must equal a full decompress."""
+    sp = TopKCompressor(0.01)
+    wc = IndexCompressor(sp, {"index": "bloom", "policy": "leftmost"})
+    payload, ctx = wc.compress(grad, "w")
+    own = wc.decompress_own(payload, ctx, "w")  # consumes cache
+    full = wc.decompress(payload, ctx)
+    assert torch.allclose(own, full)
+
+    wb = DeepReduce(sp, {"value": "polyfit", "index": "bloom"})
+    payload, ctx = wb.compress(grad, "w")
+    own = wb.decompress_own(payload, ctx, "w")
+    full = wb.decompress(payload, ctx)
+    assert torch.allclose(own, full)
+
+
+def test_residual_uses_own_decode(grad):
+    from deepreduce_amd import ResidualMemory
+
+    sp = TopKCompressor(0.01)
+    wc = IndexCompressor(sp, {"index": "bloom", "policy": "leftmost"})
+    mem = ResidualMemory()
+    t = mem.compensate(grad, "w")
+    payload, ctx = wc.compress(t, "w")
+    mem.update(t, "w", wc, payload, ctx)
+    # residual + own-decompressed == compensated
+    assert torch.allclose(mem.residuals["w"] + wc.decompress(payload, ctx), t, atol=1e-6)

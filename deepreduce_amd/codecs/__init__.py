@@ -33,6 +33,7 @@ from .bloom import Bloom, BloomCPU  # noqa: E402
 from .polyfit import PolyFit  # noqa: E402
 from .polyfit_cpu import PolyFitCPU  # noqa: E402
 from .doubleexp import DoubleExp  # noqa: E402
+from .polyseg import PolySeg  # noqa: E402
 from .qsgd import QSGD  # noqa: E402
 from .rle import RunLength  # noqa: E402
 from .gzipc import Gzip  # noqa: E402
@@ -45,6 +46,7 @@ compressor = {
     "bloom_cpu": BloomCPU,
     "polyfit_cpu": PolyFitCPU,
     "doubleexp": DoubleExp,
+    "polyseg": PolySeg,
     "gzip": Gzip,
     "huffman": Huffman,
     "rle": RunLength,

@@ -96,8 +96,12 @@ class Allgather(Communicator):
         buffer, metas = _flatten_payload(tensors)
         self.last_wire_bytes = buffer.numel()
         if world == 1:
-            decompressed = self.compressor.decompress(tensors, ctx)
-            return decompressed if not self.compressor.average else decompressed
+            # single-rank: reuse the compress-side decode cache when the
+            # wrapper provides it (skips e.g. a second bloom universe query)
+            own = getattr(self.compressor, "decompress_own", None)
+            if own is not None:
+                return own(tensors, ctx, name)
+            return self.compressor.decompress(tensors, ctx)
 
         if self.compressor.tensors_size_are_same:
             gathered = [torch.empty_like(buffer) for _ in range(world)]
@@ -107,9 +111,13 @@ class Allgather(Communicator):
             payloads = self._ragged_gather(buffer, metas, world)
 
         total = None
-        for p in payloads:
-            d = self.compressor.decompress(p, ctx)
-            total = d if total is None else total + d
+        batch = getattr(self.compressor, "decompress_batch", None)
+        if batch is not None:
+            total = batch(payloads, ctx)  # fused multi-rank path (or None)
+        if total is None:
+            for p in payloads:
+                d = self.compressor.decompress(p, ctx)
+                total = d if total is None else total + d
         if self.compressor.average:
             total = total / world
         return total

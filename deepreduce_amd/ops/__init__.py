@@ -72,6 +72,20 @@ def bloom_query_positives(packed: torch.Tensor, m: int, num_hash: int, universe:
     return _ref.bloom_query_positives(packed, m, num_hash, universe)
 
 
+def bloom_query_positives_multi(packed2d: torch.Tensor, m: int, num_hash: int, universe: int):
+    """R stacked filters (same m/k/universe) -> (positives concatenated
+    rank-major, per-rank counts int64[R]).  On GPU the hash computation is
+    amortized across ranks (the probe positions are filter-independent)."""
+    if _want_hip(packed2d):
+        return _hip.bloom_query_positives_multi(packed2d, m, num_hash, universe)
+    outs = [
+        _ref.bloom_query_positives(packed2d[r].contiguous(), m, num_hash, universe)
+        for r in range(packed2d.shape[0])
+    ]
+    counts = torch.tensor([o.numel() for o in outs], dtype=torch.int64)
+    return torch.cat(outs) if outs else torch.empty(0, dtype=torch.int64), counts
+
+
 def bloom_query_members(packed: torch.Tensor, m: int, num_hash: int, items: torch.Tensor):
     if _want_hip(packed):
         return _hip.bloom_query_members(packed, m, num_hash, items)

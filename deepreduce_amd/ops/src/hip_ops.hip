@@ -44,15 +44,21 @@ __device__ __forceinline__ void hash_bases(int64_t item, uint32_t* h1, uint32_t*
     *h2 = fmix32(*h1 ^ H2_SALT) | 1u;
 }
 
+// Incremental double hashing: (h1 + j*h2) mod m == iterate pos += (h2 mod m)
+// with one conditional subtract — bit-identical to hashing.py's int64 mod,
+// without a 64-bit mod per probe.
 __device__ __forceinline__ bool bloom_test(const uint8_t* __restrict__ bits, int64_t m,
                                            int k, int64_t item) {
     uint32_t h1, h2;
     hash_bases(item, &h1, &h2);
-    for (int j = 0; j < k; ++j) {
-        int64_t pos = (int64_t)(((uint64_t)h1 + (uint64_t)j * h2) % (uint64_t)m);
+    uint64_t pos = (uint64_t)h1 % (uint64_t)m;
+    uint64_t step = (uint64_t)h2 % (uint64_t)m;
+    for (int j = 0;;) {
         if (!((bits[pos >> 3] >> (pos & 7)) & 1)) return false;
+        if (++j >= k) return true;
+        pos += step;
+        if (pos >= (uint64_t)m) pos -= (uint64_t)m;
     }
-    return true;
 }
 
 // ---------------------------------------------------------------------------
@@ -97,78 +103,163 @@ torch::Tensor bloom_insert(torch::Tensor idxs, int64_t m, int64_t num_hash) {
 // ---------------------------------------------------------------------------
 
 #define QBLOCK 256
-#define QCHUNK (QBLOCK * 32)  // items per block
+#define MAXR 16  // max ranks per batched query
 
-__global__ void bloom_count_kernel(const uint8_t* __restrict__ bits, int64_t m, int k,
-                                   int64_t universe, int* __restrict__ block_counts) {
-    int64_t start = (int64_t)blockIdx.x * QCHUNK;
-    int64_t end = min(start + (int64_t)QCHUNK, universe);
-    int cnt = 0;
-    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
-        cnt += bloom_test(bits, m, k, i) ? 1 : 0;
-    // wave reduce then LDS
-    for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off, WAVE);
-    __shared__ int wsum[QBLOCK / WAVE];
+// chunk sizing: enough blocks to fill 256 CUs several times over, chunk a
+// multiple of the block size
+static inline int64_t query_chunk(int64_t universe) {
+    int64_t chunk = ceil_div(universe, 2048);
+    chunk = ceil_div(chunk, QBLOCK) * QBLOCK;
+    return std::max<int64_t>(chunk, QBLOCK);
+}
+
+// Pass 1: test the filter(s), emit (a) per-block per-rank counts and (b) a
+// predicate bit-plane per rank (one uint64 ballot word per wave) so pass 2
+// never re-hashes.  R filters share the SAME probe positions (the hash does
+// not depend on the filter), so hashing is amortized R-fold.
+__global__ void bloom_count_kernel(const uint8_t* __restrict__ bits, int64_t stride_bytes,
+                                   int R, int64_t m, int k, int64_t universe, int64_t chunk,
+                                   int* __restrict__ block_counts /*[R, nblocks]*/,
+                                   uint64_t* __restrict__ mask /*[R, ceil(u/64)]*/,
+                                   int64_t mask_stride) {
+    int64_t start = (int64_t)blockIdx.x * chunk;
+    int64_t end = min(start + chunk, universe);
     int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
-    if (lane == 0) wsum[wid] = cnt;
+    int cnt[MAXR];
+    for (int r = 0; r < R; ++r) cnt[r] = 0;
+
+    for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        unsigned alive = (i < end) ? ((1u << R) - 1) : 0u;
+        if (alive) {
+            uint32_t h1, h2;
+            hash_bases(i, &h1, &h2);
+            uint64_t pos = (uint64_t)h1 % (uint64_t)m;
+            uint64_t step = (uint64_t)h2 % (uint64_t)m;
+            for (int j = 0; j < k && alive; ++j) {
+                int64_t byte = pos >> 3;
+                uint8_t bit = pos & 7;
+                for (int r = 0; r < R; ++r)
+                    if (alive & (1u << r))
+                        if (!((bits[r * stride_bytes + byte] >> bit) & 1)) alive &= ~(1u << r);
+                pos += step;
+                if (pos >= (uint64_t)m) pos -= (uint64_t)m;
+            }
+        }
+        for (int r = 0; r < R; ++r) {
+            bool pred = alive & (1u << r);
+            uint64_t ball = __ballot(pred);
+            cnt[r] += __popcll(ball);
+            if (lane == 0) mask[r * mask_stride + ((i0 + (int64_t)wid * WAVE) >> 6)] = ball;
+        }
+    }
+    // reduce counts across the block (cnt[r] is wave-uniform after popcll)
+    __shared__ int wsum[MAXR][QBLOCK / WAVE];
+    if (lane == 0)
+        for (int r = 0; r < R; ++r) wsum[r][wid] = cnt[r];
     __syncthreads();
     if (threadIdx.x == 0) {
-        int total = 0;
-        for (int w = 0; w < QBLOCK / WAVE; ++w) total += wsum[w];
-        block_counts[blockIdx.x] = total;
+        for (int r = 0; r < R; ++r) {
+            int total = 0;
+            for (int w = 0; w < QBLOCK / WAVE; ++w) total += wsum[r][w];
+            block_counts[r * gridDim.x + blockIdx.x] = total;
+        }
     }
 }
 
-__global__ void bloom_scatter_kernel(const uint8_t* __restrict__ bits, int64_t m, int k,
-                                     int64_t universe, const int* __restrict__ block_offsets,
+// Pass 2: ordered compaction of the predicate bit-plane (no hashing).
+__global__ void bloom_scatter_kernel(const uint64_t* __restrict__ mask, int64_t mask_stride,
+                                     int R, int64_t universe, int64_t chunk,
+                                     const int* __restrict__ block_offsets /*[R, nblocks]*/,
+                                     const int64_t* __restrict__ rank_base /*[R]*/,
                                      int64_t* __restrict__ out) {
-    int64_t start = (int64_t)blockIdx.x * QCHUNK;
-    int64_t end = min(start + (int64_t)QCHUNK, universe);
-    __shared__ int wave_cnt[QBLOCK / WAVE];
-    __shared__ int base_s;
-    if (threadIdx.x == 0) base_s = block_offsets[blockIdx.x];
+    int64_t start = (int64_t)blockIdx.x * chunk;
+    int64_t end = min(start + chunk, universe);
     int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    __shared__ int wave_cnt[MAXR][QBLOCK / WAVE];
+    __shared__ int base_s[MAXR];
+    if (threadIdx.x < MAXR && threadIdx.x < R)
+        base_s[threadIdx.x] =
+            block_offsets[threadIdx.x * gridDim.x + blockIdx.x] +
+            (int)rank_base[threadIdx.x];
+    __syncthreads();
     for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
         int64_t i = i0 + threadIdx.x;
-        bool pred = (i < end) && bloom_test(bits, m, k, i);
-        uint64_t ball = __ballot(pred);
-        if (lane == 0) wave_cnt[wid] = __popcll(ball);
-        __syncthreads();
-        int wbase = 0, total = 0;
-        for (int w = 0; w < QBLOCK / WAVE; ++w) {
-            if (w < wid) wbase += wave_cnt[w];
-            total += wave_cnt[w];
-        }
-        if (pred) {
-            int prefix = __popcll(ball & ((lane == 63) ? ~0ull >> 1 : ((1ull << lane) - 1)));
-            out[base_s + wbase + prefix] = i;
+        int64_t w64 = (i0 + (int64_t)wid * WAVE) >> 6;
+        for (int r = 0; r < R; ++r) {
+            uint64_t ball = mask[r * mask_stride + w64];
+            if (lane == 0) wave_cnt[r][wid] = __popcll(ball);
         }
         __syncthreads();
-        if (threadIdx.x == 0) base_s += total;
+        for (int r = 0; r < R; ++r) {
+            uint64_t ball = mask[r * mask_stride + w64];
+            bool pred = (ball >> lane) & 1;
+            if (pred) {
+                int wbase = 0;
+                for (int w = 0; w < wid; ++w) wbase += wave_cnt[r][w];
+                int prefix = __popcll(ball & ((lane == 63) ? ~0ull >> 1 : ((1ull << lane) - 1)));
+                out[base_s[r] + wbase + prefix] = i;
+            }
+        }
+        __syncthreads();
+        if (threadIdx.x < MAXR && threadIdx.x < R) {
+            int total = 0;
+            for (int w = 0; w < QBLOCK / WAVE; ++w) total += wave_cnt[threadIdx.x][w];
+            base_s[threadIdx.x] += total;
+        }
         __syncthreads();
     }
+}
+
+// shared driver: R stacked filters -> (positives flat [sum], counts [R])
+static std::vector<torch::Tensor> query_multi_impl(torch::Tensor bits2d, int64_t m,
+                                                   int64_t num_hash, int64_t universe) {
+    auto bits = bits2d.contiguous();
+    int R = (int)bits.size(0);
+    TORCH_CHECK(R >= 1 && R <= MAXR, "1..16 ranks supported");
+    int64_t stride_bytes = bits.size(1);
+    int64_t chunk = query_chunk(universe);
+    int64_t nblocks = ceil_div(universe, chunk);
+    auto dev = bits.device();
+    auto counts = torch::empty({R, nblocks}, torch::dtype(torch::kInt32).device(dev));
+    // mask rows padded so every wave's ballot word has a slot
+    int64_t mask_words = ceil_div(nblocks * chunk, 64);
+    auto mask = torch::empty({R, mask_words}, torch::dtype(torch::kInt64).device(dev));
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(bloom_count_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
+                       bits.data_ptr<uint8_t>(), stride_bytes, R, m, (int)num_hash, universe,
+                       chunk, counts.data_ptr<int>(), (uint64_t*)mask.data_ptr<int64_t>(),
+                       mask_words);
+    auto csum = counts.cumsum(1, torch::kInt32);
+    auto offsets = (csum - counts).to(torch::kInt32).contiguous();
+    auto rank_totals = csum.select(1, nblocks - 1).to(torch::kInt64);
+    auto rank_base = rank_totals.cumsum(0) - rank_totals;
+    int64_t total = (int64_t)rank_totals.sum().item<int64_t>();  // one sync
+    auto out = torch::empty({total}, torch::dtype(torch::kInt64).device(dev));
+    if (total > 0) {
+        auto rank_base_c = rank_base.contiguous();
+        hipLaunchKernelGGL(bloom_scatter_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
+                           (const uint64_t*)mask.data_ptr<int64_t>(), mask_words, R, universe,
+                           chunk, offsets.data_ptr<int>(), rank_base_c.data_ptr<int64_t>(),
+                           out.data_ptr<int64_t>());
+    }
+    return {out, rank_totals};
 }
 
 torch::Tensor bloom_query_positives(torch::Tensor packed, int64_t m, int64_t num_hash,
                                     int64_t universe) {
     CHECK_CUDA(packed);
-    auto bits = packed.contiguous();
-    int64_t nblocks = ceil_div(universe, QCHUNK);
-    auto counts = torch::empty({nblocks}, torch::dtype(torch::kInt32).device(bits.device()));
-    hipStream_t stream = at::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(bloom_count_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
-                       bits.data_ptr<uint8_t>(), m, (int)num_hash, universe,
-                       counts.data_ptr<int>());
-    auto csum = counts.cumsum(0, torch::kInt32);
-    auto offsets = (csum - counts).to(torch::kInt32);
-    int64_t total = csum.numel() ? csum[-1].item<int64_t>() : 0;  // one sync (output size)
-    auto out = torch::empty({total}, torch::dtype(torch::kInt64).device(bits.device()));
-    if (total > 0) {
-        hipLaunchKernelGGL(bloom_scatter_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
-                           bits.data_ptr<uint8_t>(), m, (int)num_hash, universe,
-                           offsets.data_ptr<int>(), out.data_ptr<int64_t>());
-    }
-    return out;
+    auto res = query_multi_impl(packed.contiguous().unsqueeze(0), m, num_hash, universe);
+    return res[0];
+}
+
+// batched: packed2d [R, nbytes] (one row per rank, same m/k/universe) ->
+// (positives concatenated rank-major, per-rank counts)
+std::vector<torch::Tensor> bloom_query_positives_multi(torch::Tensor packed2d, int64_t m,
+                                                       int64_t num_hash, int64_t universe) {
+    CHECK_CUDA(packed2d);
+    TORCH_CHECK(packed2d.dim() == 2, "expected [R, nbytes]");
+    return query_multi_impl(packed2d, m, num_hash, universe);
 }
 
 __global__ void bloom_members_kernel(const uint8_t* __restrict__ bits, int64_t m, int k,
@@ -361,6 +452,8 @@ torch::Tensor unpack_ints(torch::Tensor stream, int64_t n, int64_t nbits) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bloom_insert", &bloom_insert, "Bloom insert (HIP)");
     m.def("bloom_query_positives", &bloom_query_positives, "Bloom full-universe query (HIP)");
+    m.def("bloom_query_positives_multi", &bloom_query_positives_multi,
+          "Batched multi-rank Bloom query (HIP): hash once, test R filters");
     m.def("bloom_query_members", &bloom_query_members, "Bloom membership test (HIP)");
     m.def("qsgd_quantize", &qsgd_quantize, "QSGD quantize (HIP)");
     m.def("qsgd_dequantize", &qsgd_dequantize, "QSGD dequantize (HIP)");

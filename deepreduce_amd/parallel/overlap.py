@@ -151,9 +151,13 @@ class OverlappedReducer:
                     else:
                         payloads = [_unflatten_payload(b, metas) for b in gathered]
                     total = None
-                    for pay in payloads:
-                        d = grc.compressor.decompress(pay, ctx)
-                        total = d if total is None else total + d
+                    batch = getattr(grc.compressor, "decompress_batch", None)
+                    if batch is not None:
+                        total = batch(payloads, ctx)
+                    if total is None:
+                        for pay in payloads:
+                            d = grc.compressor.decompress(pay, ctx)
+                            total = d if total is None else total + d
                     out = total / world if grc.compressor.average else total
                 p_.grad.data.copy_(out.view_as(p_.grad.data))
             self._exchange_small(world)

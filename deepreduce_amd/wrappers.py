@@ -47,7 +47,7 @@ class _WrapperBase(Compressor):
         return self.sparsifier.aggregate(tensors)
 
     def decompress_own(self, tensors, ctx, name):
-        cached = self._own_cache.pop(name, None)
+        cached = self._own_cache.get(name)
         if cached is not None:
             vals, idxs, shape = cached
             return self.sparsifier.decompress((vals, idxs), shape)
@@ -70,8 +70,9 @@ class ValueCompressor(_WrapperBase):
         super().__init__(sparsifier, params)
         name = self.params.get("value", "polyfit")
         self.val_codec = codec_registry[name]
-        if name not in ("qsgd",):
-            # polyfit coefficient count varies with per-rank num_pos
+        if name not in ("qsgd", "polyseg"):
+            # polyfit coefficient count varies with per-rank num_pos;
+            # qsgd and polyseg payload sizes depend only on k -> uniform
             self.tensors_size_are_same = False
 
     def compress(self, tensor, name):
@@ -146,6 +147,45 @@ class IndexCompressor(_WrapperBase):
                 print(f"val_relative_volume: {tensor_bits([tensors[0]]) / dense_bits:.4f}")
         return self.sparsifier.decompress((vals, idxs), shape)
 
+    def decompress_batch(self, payloads, ctx):
+        """Fused multi-rank decompress (bloom + leftmost only): ONE batched
+        universe query over all ranks' filters — the probe positions are
+        filter-independent, so hashing amortizes R-fold on the GPU — then a
+        single fused scatter-add.  Returns the SUM of dense tensors, or
+        None if this codec/policy combination has no fast path.
+        """
+        from .codecs.bloom import Bloom
+
+        shape = ctx
+        numel = int(torch.Size(shape).numel())
+        if (
+            self.idx_codec is not Bloom
+            or self.params.get("policy", "leftmost") != "leftmost"
+            or numel <= _BYPASS_NUMEL
+            or len(payloads) < 2
+        ):
+            return None
+        from . import ops
+
+        vals0 = payloads[0][0]
+        num_indices = int(vals0.numel())
+        if any(int(p[0].numel()) != num_indices for p in payloads):
+            return None  # leftmost assumes uniform k (topk sparsifier)
+        num_hash, m = Bloom._config(num_indices, numel, self.params)
+        bits = torch.stack([p[1].contiguous() for p in payloads])
+        pos, counts = ops.bloom_query_positives_multi(bits, m, num_hash, numel)
+        counts_l = counts.tolist()  # one small sync for slicing
+        base = 0
+        idx_parts, val_parts = [], []
+        for r, c in enumerate(counts_l):
+            n = min(int(c), num_indices)
+            idx_parts.append(pos[base : base + n])
+            val_parts.append(payloads[r][0][:n])
+            base += int(c)
+        dense = torch.zeros(numel, dtype=vals0.dtype, device=vals0.device)
+        dense.index_add_(0, torch.cat(idx_parts), torch.cat(val_parts))
+        return dense.view(shape)
+
 
 class DeepReduce(_WrapperBase):
     """Joint index+value compression glued by a mapping permutation.
@@ -184,7 +224,16 @@ class DeepReduce(_WrapperBase):
             new_idxs = torch.arange(vals.numel(), device=vals.device)
             vals, mapping, shape_out = self.val_codec.compress((vals, new_idxs, shape), self.params)
             ctx = shape_out
-            tensors = (vals, idxs, mapping.int())
+            if self.params.get("pack_mapping", True):
+                # ceil(log2 k) bits per mapping entry instead of int32
+                # (paper App. E; the reference left this commented out at
+                # pytorch/deepreduce.py:264-265 — we ship it, GPU-packed)
+                from .codecs.intpack import pack_with_header
+
+                mapping = pack_with_header(mapping.long())
+                tensors = (vals, idxs, mapping)
+            else:
+                tensors = (vals, idxs, mapping.int())
         if self.params.get("micro-benchmark", False):
             _sync_if(tensor)
             print(f"_compression time:{time.perf_counter() - start}")
@@ -195,7 +244,7 @@ class DeepReduce(_WrapperBase):
         start = time.perf_counter()
         if torch.Size(shape).numel() > _BYPASS_NUMEL:
             vals, idxs, mapping = tensors
-            mapping = mapping.long()
+            mapping = self._unpack_mapping(mapping)
             vals, _, _ = self.val_codec.decompress((vals, mapping, shape), self.params)
             _, idxs, _ = self.idx_codec.decompress((mapping, idxs, shape), self.params)
             idxs = idxs[mapping]
@@ -208,15 +257,22 @@ class DeepReduce(_WrapperBase):
             print(f"val_relative_volume: {tensor_bits([tensors[0]]) / dense_bits:.4f}")
         return self.sparsifier.decompress((vals, idxs), shape)
 
+    def _unpack_mapping(self, mapping):
+        if mapping.dtype == torch.uint8:  # bit-packed wire
+            from .codecs.intpack import unpack_with_header
+
+            return unpack_with_header(mapping).to(mapping.device)
+        return mapping.long()
+
     def decompress_own(self, tensors, ctx, name):
         """Own-payload decode with the cached bloom indices: only the (cheap)
         value-codec eval runs; the full-universe query is skipped."""
-        cached_idxs = self._own_cache.pop(name, None)
+        cached_idxs = self._own_cache.get(name)
         shape = ctx
         if cached_idxs is None or torch.Size(shape).numel() <= _BYPASS_NUMEL:
             return self.decompress(tensors, ctx)
         vals, idxs, mapping = tensors
-        mapping = mapping.long()
+        mapping = self._unpack_mapping(mapping)
         vals, _, _ = self.val_codec.decompress((vals, mapping, shape), self.params)
         idxs = cached_idxs[mapping]
         return self.sparsifier.decompress((vals, idxs), shape)

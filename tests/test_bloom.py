@@ -119,3 +119,40 @@ def test_wire_volume_beats_raw_indices(data):
     params = {"policy": "leftmost", "fpr": 0.01}
     v, bits, shape = compressor["bloom"].compress((vals, idxs, t.size()), params)
     assert bits.numel() < idxs.numel() * 4  # paper: ~50% of int32 keys
+
+
+def test_decompress_batch_matches_loop(data):
+    """the fused multi-rank decompress must equal the per-rank loop sum."""
+    from deepreduce_amd import TopKCompressor
+    from deepreduce_amd.wrappers import IndexCompressor
+
+    t, vals, idxs = data
+    sp = TopKCompressor(0.01)
+    wc = IndexCompressor(sp, {"index": "bloom", "policy": "leftmost"})
+    # three 'ranks' with different gradients
+    payloads, ctx = [], None
+    for r in range(3):
+        torch.manual_seed(500 + r)
+        g = torch.randn(t.numel())
+        payload, ctx = wc.compress(g, f"w")
+        payloads.append(payload)
+    fused = wc.decompress_batch(payloads, ctx)
+    assert fused is not None
+    loop = sum(wc.decompress(p, ctx) for p in payloads)
+    assert torch.allclose(fused, loop)
+
+
+def test_query_multi_cpu_fallback(data):
+    from deepreduce_amd.ops import bloom_insert, bloom_query_positives, bloom_query_positives_multi
+
+    t, vals, idxs = data
+    nh, m = 8, 200_003
+    b1 = bloom_insert(idxs, m, nh)
+    b2 = bloom_insert(idxs + 1, m, nh)
+    stacked = torch.stack([b1, b2])
+    pos, counts = bloom_query_positives_multi(stacked, m, nh, t.numel())
+    p1 = bloom_query_positives(b1, m, nh, t.numel())
+    p2 = bloom_query_positives(b2, m, nh, t.numel())
+    assert counts.tolist() == [p1.numel(), p2.numel()]
+    assert torch.equal(pos[: p1.numel()], p1)
+    assert torch.equal(pos[p1.numel() :], p2)

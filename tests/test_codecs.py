@@ -147,3 +147,24 @@ class TestLossy:
             assert (val >= 0) == (orig_by_idx[ix] >= 0) or abs(orig_by_idx[ix]) < 1e-6
         rel = (v2.abs().sort().values - vals.abs().sort().values).norm() / vals.norm()
         assert rel < 0.1
+
+
+class TestPolySeg:
+    def test_roundtrip_and_uniform_payload(self, sparse):
+        t, vals, idxs = sparse
+        v, i, shape = compressor["polyseg"].compress((vals, idxs, t.size()), {})
+        v2, i2, _ = compressor["polyseg"].decompress((v, i, shape), {})
+        sorted_desc = vals.sort(descending=True).values
+        rel = (v2 - sorted_desc).norm() / sorted_desc.norm()
+        assert rel < 0.1
+        # payload size depends only on k, not on the value distribution
+        vals_b = -vals.abs()  # all negative: different num_pos entirely
+        vb, _, _ = compressor["polyseg"].compress((vals_b, idxs, t.size()), {})
+        assert vb.numel() == v.numel()
+
+    def test_fixed_segments_deterministic(self):
+        from deepreduce_amd.codecs.polyseg import fixed_segments
+
+        assert fixed_segments(500, 10) == fixed_segments(500, 10)
+        assert sum(fixed_segments(12345, 10)) == 12345
+        assert sum(fixed_segments(7, 10)) == 7

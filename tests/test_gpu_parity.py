@@ -118,10 +118,31 @@ def test_codec_end_to_end_gpu(dev):
     assert torch.allclose(t[i2], v2)
     true = set(idxs.cpu().tolist())
     rec = set(i2.cpu().tolist())
-    assert len(true & rec) / len(true) > 0.95
+    # leftmost policy: expected recall ~ 1 - fpr*d/k = 0.9 here
+    assert len(true & rec) / len(true) > 0.85
 
 
 def test_smoke_entrypoint():
     import __graft_entry__ as ge
 
     ge.smoke()
+
+
+def test_bloom_query_multi_parity(hip, dev):
+    """batched R-filter query == R independent queries, on GPU."""
+    torch.manual_seed(9)
+    universe = 1_500_000
+    m, k = 400_009, 9
+    filters, singles = [], []
+    for r in range(8):
+        idxs = torch.randperm(universe, device=dev)[:15_000]
+        b = hip.bloom_insert(idxs, m, k)
+        filters.append(b)
+        singles.append(hip.bloom_query_positives(b, m, k, universe))
+    pos, counts = hip.bloom_query_positives_multi(torch.stack(filters), m, k, universe)
+    base = 0
+    for r in range(8):
+        n = int(counts[r])
+        assert n == singles[r].numel(), f"rank {r}"
+        assert torch.equal(pos[base : base + n], singles[r]), f"rank {r}"
+        base += n

@@ -73,7 +73,9 @@ def test_both_mode_mapping_roundtrip(grad):
     wc = DeepReduce(sp, {"value": "polyfit", "index": "bloom", "policy": "leftmost"})
     payload, ctx = wc.compress(grad, "w")
     vals, bits, mapping = payload
-    assert mapping.dtype == torch.int32
+    assert mapping.dtype == torch.uint8  # bit-packed (paper App. E)
+    k = grad.numel() // 100
+    assert mapping.numel() < k * 4  # beats the reference's int32 mapping
     assert vals.dtype == torch.float64
     out = wc.decompress(payload, ctx)
     assert out.shape == grad.shape

@@ -60,15 +60,24 @@ def topk_select(flat: torch.Tensor, k: int):
     return _ref.topk_select(flat, k)
 
 
+def _cpu_native() -> bool:
+    """C++ CPU paths (same .so) are used opportunistically when loadable."""
+    return _load_hip() is not None and os.environ.get("DEEPREDUCE_FORCE_TORCH_CPU") != "1"
+
+
 def bloom_insert(idxs: torch.Tensor, m: int, num_hash: int) -> torch.Tensor:
     if _want_hip(idxs):
         return _hip.bloom_insert(idxs, m, num_hash)
+    if _cpu_native():
+        return _hip.bloom_insert_cpu(idxs, m, num_hash)
     return _ref.bloom_insert(idxs, m, num_hash)
 
 
 def bloom_query_positives(packed: torch.Tensor, m: int, num_hash: int, universe: int):
     if _want_hip(packed):
         return _hip.bloom_query_positives(packed, m, num_hash, universe)
+    if _cpu_native():
+        return _hip.bloom_query_positives_cpu(packed, m, num_hash, universe)
     return _ref.bloom_query_positives(packed, m, num_hash, universe)
 
 
@@ -89,18 +98,24 @@ def bloom_query_positives_multi(packed2d: torch.Tensor, m: int, num_hash: int, u
 def bloom_query_members(packed: torch.Tensor, m: int, num_hash: int, items: torch.Tensor):
     if _want_hip(packed):
         return _hip.bloom_query_members(packed, m, num_hash, items)
+    if _cpu_native():
+        return _hip.bloom_query_members_cpu(packed, m, num_hash, items)
     return _ref.bloom_query_members(packed, m, num_hash, items)
 
 
 def pack_ints(values: torch.Tensor, nbits: int) -> torch.Tensor:
     if _want_hip(values):
         return _hip.pack_ints(values, nbits)
+    if _cpu_native():
+        return _hip.pack_ints_cpu(values, nbits)
     return _ref.pack_ints(values, nbits)
 
 
 def unpack_ints(stream: torch.Tensor, n: int, nbits: int) -> torch.Tensor:
     if _want_hip(stream):
         return _hip.unpack_ints(stream, n, nbits)
+    if _cpu_native():
+        return _hip.unpack_ints_cpu(stream, n, nbits)
     return _ref.unpack_ints(stream, n, nbits)
 
 

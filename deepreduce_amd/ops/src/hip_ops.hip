@@ -27,7 +27,7 @@ static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // hashing (must mirror deepreduce_amd/hashing.py exactly)
 // ---------------------------------------------------------------------------
 
-__device__ __forceinline__ uint32_t fmix32(uint32_t h) {
+__host__ __device__ __forceinline__ uint32_t fmix32(uint32_t h) {
     h ^= h >> 16;
     h *= 0x85EBCA6Bu;
     h ^= h >> 13;
@@ -38,7 +38,7 @@ __device__ __forceinline__ uint32_t fmix32(uint32_t h) {
 
 #define H2_SALT 0x6B43A9B5u
 
-__device__ __forceinline__ void hash_bases(int64_t item, uint32_t* h1, uint32_t* h2) {
+__host__ __device__ __forceinline__ void hash_bases(int64_t item, uint32_t* h1, uint32_t* h2) {
     uint32_t x = (uint32_t)(item & 0xFFFFFFFFll);
     *h1 = fmix32(x + 1u);
     *h2 = fmix32(*h1 ^ H2_SALT) | 1u;
@@ -47,7 +47,7 @@ __device__ __forceinline__ void hash_bases(int64_t item, uint32_t* h1, uint32_t*
 // Incremental double hashing: (h1 + j*h2) mod m == iterate pos += (h2 mod m)
 // with one conditional subtract — bit-identical to hashing.py's int64 mod,
 // without a 64-bit mod per probe.
-__device__ __forceinline__ bool bloom_test(const uint8_t* __restrict__ bits, int64_t m,
+__host__ __device__ __forceinline__ bool bloom_test(const uint8_t* __restrict__ bits, int64_t m,
                                            int k, int64_t item) {
     uint32_t h1, h2;
     hash_bases(item, &h1, &h2);
@@ -447,6 +447,114 @@ torch::Tensor unpack_ints(torch::Tensor stream, int64_t n, int64_t nbits) {
     return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// CPU-native C++ paths (replace the reference's TF C++ CPU ops:
+// bloom_filter_compression.cc / integer_compression.cc) — same wire format
+// and hash math as the HIP kernels, parallelized with at::parallel_for.
+// ---------------------------------------------------------------------------
+
+#include <ATen/Parallel.h>
+
+torch::Tensor bloom_insert_cpu(torch::Tensor idxs, int64_t m, int64_t num_hash) {
+    auto items = idxs.to(torch::kInt64).contiguous();
+    int64_t nbytes = ceil_div(m, 8);
+    auto out = torch::zeros({nbytes}, torch::dtype(torch::kUInt8));
+    uint8_t* bits = out.data_ptr<uint8_t>();
+    const int64_t* it = items.data_ptr<int64_t>();
+    int64_t n = items.numel();
+    // serial (insert is tiny: k items, k*num_hash bit sets)
+    for (int64_t i = 0; i < n; ++i) {
+        uint32_t h1, h2;
+        hash_bases(it[i], &h1, &h2);
+        uint64_t pos = (uint64_t)h1 % (uint64_t)m;
+        uint64_t step = (uint64_t)h2 % (uint64_t)m;
+        for (int64_t j = 0; j < num_hash; ++j) {
+            bits[pos >> 3] |= (uint8_t)(1u << (pos & 7));
+            pos += step;
+            if (pos >= (uint64_t)m) pos -= (uint64_t)m;
+        }
+    }
+    return out;
+}
+
+torch::Tensor bloom_query_positives_cpu(torch::Tensor packed, int64_t m, int64_t num_hash,
+                                        int64_t universe) {
+    auto p = packed.contiguous();
+    const uint8_t* bits = p.data_ptr<uint8_t>();
+    const int64_t chunk = 1 << 16;
+    int64_t nchunks = ceil_div(universe, chunk);
+    std::vector<std::vector<int64_t>> found((size_t)nchunks);
+    at::parallel_for(0, nchunks, 1, [&](int64_t c0, int64_t c1) {
+        for (int64_t c = c0; c < c1; ++c) {
+            int64_t start = c * chunk, end = std::min(start + chunk, universe);
+            auto& v = found[(size_t)c];
+            for (int64_t i = start; i < end; ++i)
+                if (bloom_test(bits, m, (int)num_hash, i)) v.push_back(i);
+        }
+    });
+    int64_t total = 0;
+    for (auto& v : found) total += (int64_t)v.size();
+    auto out = torch::empty({total}, torch::dtype(torch::kInt64));
+    int64_t* o = out.data_ptr<int64_t>();
+    for (auto& v : found) {
+        std::copy(v.begin(), v.end(), o);
+        o += v.size();
+    }
+    return out;
+}
+
+torch::Tensor bloom_query_members_cpu(torch::Tensor packed, int64_t m, int64_t num_hash,
+                                      torch::Tensor items) {
+    auto p = packed.contiguous();
+    auto it = items.to(torch::kInt64).contiguous();
+    const uint8_t* bits = p.data_ptr<uint8_t>();
+    const int64_t* iv = it.data_ptr<int64_t>();
+    int64_t n = it.numel();
+    auto out = torch::empty({n}, torch::dtype(torch::kBool));
+    bool* o = out.data_ptr<bool>();
+    at::parallel_for(0, n, 4096, [&](int64_t a, int64_t b) {
+        for (int64_t i = a; i < b; ++i) o[i] = bloom_test(bits, m, (int)num_hash, iv[i]);
+    });
+    return out;
+}
+
+torch::Tensor pack_ints_cpu(torch::Tensor values, int64_t nbits) {
+    auto v = values.to(torch::kInt64).contiguous();
+    int64_t n = v.numel();
+    int64_t nbytes = ceil_div(n * nbits, 8);
+    auto out = torch::zeros({nbytes}, torch::dtype(torch::kUInt8));
+    uint8_t* o = out.data_ptr<uint8_t>();
+    const int64_t* vv = v.data_ptr<int64_t>();
+    for (int64_t i = 0; i < n; ++i) {
+        int64_t bit0 = i * nbits;
+        uint64_t val = (uint64_t)vv[i];
+        for (int64_t b = 0; b < nbits; ++b) {
+            int64_t bit = bit0 + b;
+            o[bit >> 3] |= (uint8_t)(((val >> b) & 1) << (bit & 7));
+        }
+    }
+    return out;
+}
+
+torch::Tensor unpack_ints_cpu(torch::Tensor stream, int64_t n, int64_t nbits) {
+    auto s = stream.contiguous();
+    const uint8_t* sv = s.data_ptr<uint8_t>();
+    auto out = torch::empty({n}, torch::dtype(torch::kInt64));
+    int64_t* o = out.data_ptr<int64_t>();
+    at::parallel_for(0, n, 8192, [&](int64_t a, int64_t b) {
+        for (int64_t i = a; i < b; ++i) {
+            int64_t bit0 = i * nbits;
+            uint64_t acc = 0;
+            int need = (int)((nbits + (bit0 & 7) + 7) / 8);
+            for (int k = 0; k < need; ++k) acc |= (uint64_t)sv[(bit0 >> 3) + k] << (8 * k);
+            o[i] = (int64_t)((acc >> (bit0 & 7)) &
+                             ((nbits == 64) ? ~0ull : ((1ull << nbits) - 1)));
+        }
+    });
+    return out;
+}
+
 // ---------------------------------------------------------------------------
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -459,4 +567,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("qsgd_dequantize", &qsgd_dequantize, "QSGD dequantize (HIP)");
     m.def("pack_ints", &pack_ints, "n-bit pack (HIP)");
     m.def("unpack_ints", &unpack_ints, "n-bit unpack (HIP)");
+    m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
+    m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
+    m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");
+    m.def("pack_ints_cpu", &pack_ints_cpu, "n-bit pack (C++ CPU)");
+    m.def("unpack_ints_cpu", &unpack_ints_cpu, "n-bit unpack (C++ CPU)");
 }

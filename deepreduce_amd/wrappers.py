@@ -19,6 +19,8 @@ import time
 
 import torch
 
+from torch.profiler import record_function
+
 from .codecs import compressor as codec_registry
 from .compressors import Compressor
 from .helper import tensor_bits
@@ -81,7 +83,8 @@ class ValueCompressor(_WrapperBase):
         shape = ctx
         if torch.Size(shape).numel() > _BYPASS_NUMEL:
             start = time.perf_counter()
-            vals, idxs, shape_out = self.val_codec.compress((vals, idxs, tensor.size()), self.params)
+            with record_function("dr::val_compress"):
+                vals, idxs, shape_out = self.val_codec.compress((vals, idxs, tensor.size()), self.params)
             if self.params.get("micro-benchmark", False):
                 _sync_if(tensor)
                 print(f"val_compression time:{time.perf_counter() - start}")
@@ -93,7 +96,8 @@ class ValueCompressor(_WrapperBase):
         vals, idxs = tensors
         if torch.Size(shape).numel() > _BYPASS_NUMEL:
             start = time.perf_counter()
-            vals, idxs, shape = self.val_codec.decompress((vals, idxs, shape), self.params)
+            with record_function("dr::val_decompress"):
+                vals, idxs, shape = self.val_codec.decompress((vals, idxs, shape), self.params)
             if self.params.get("micro-benchmark", False):
                 print(f"val_decompression time:{time.perf_counter() - start}")
                 dense_bits = torch.Size(shape).numel() * 32
@@ -123,7 +127,8 @@ class IndexCompressor(_WrapperBase):
         if torch.Size(shape).numel() > _BYPASS_NUMEL:
             self.params["dense_tensor"] = tensor
             start = time.perf_counter()
-            vals, idxs, shape_out = self.idx_codec.compress((vals, idxs, tensor.size()), self.params)
+            with record_function("dr::idx_compress"):
+                vals, idxs, shape_out = self.idx_codec.compress((vals, idxs, tensor.size()), self.params)
             self.params.pop("dense_tensor", None)
             own = self.params.pop("_own_decoded", None)
             if own is not None:
@@ -139,7 +144,8 @@ class IndexCompressor(_WrapperBase):
         vals, idxs = tensors
         if torch.Size(shape).numel() > _BYPASS_NUMEL:
             start = time.perf_counter()
-            vals, idxs, shape = self.idx_codec.decompress((vals, idxs, shape), self.params)
+            with record_function("dr::idx_decompress"):
+                vals, idxs, shape = self.idx_codec.decompress((vals, idxs, shape), self.params)
             if self.params.get("micro-benchmark", False):
                 print(f"idx_decompression time:{time.perf_counter() - start}")
                 dense_bits = torch.Size(shape).numel() * 32

@@ -145,3 +145,36 @@ def test_fmix32_known_values():
     out = fmix32(xs)
     for x, o in zip(xs.tolist(), out.tolist()):
         assert o == ref(x)
+
+
+def test_cpu_native_matches_torch_reference():
+    """C++ CPU ops (if the extension is loadable) must be bit-identical to
+    the torch reference implementations."""
+    import deepreduce_amd.ops as ops
+    from deepreduce_amd.ops import reference as ref
+
+    if not ops.hip_available():
+        import pytest
+
+        pytest.skip("extension not built")
+    from deepreduce_amd import _hip_ops
+
+    idxs = torch.randperm(200_000)[:2000]
+    m, k = 40_009, 7
+    assert torch.equal(_hip_ops.bloom_insert_cpu(idxs, m, k), ref.bloom_insert(idxs, m, k))
+    packed = ref.bloom_insert(idxs, m, k)
+    assert torch.equal(
+        _hip_ops.bloom_query_positives_cpu(packed, m, k, 200_000),
+        ref.bloom_query_positives(packed, m, k, 200_000),
+    )
+    probe = torch.arange(0, 200_000, 17)
+    assert torch.equal(
+        _hip_ops.bloom_query_members_cpu(packed, m, k, probe),
+        ref.bloom_query_members(packed, m, k, probe),
+    )
+    for nbits in (1, 5, 13, 21):
+        v = torch.randint(0, 2**nbits, (5000,))
+        assert torch.equal(_hip_ops.pack_ints_cpu(v, nbits), ref.pack_ints(v, nbits))
+        assert torch.equal(
+            _hip_ops.unpack_ints_cpu(ref.pack_ints(v, nbits), 5000, nbits), v.long()
+        )

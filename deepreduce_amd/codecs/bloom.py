@@ -125,8 +125,12 @@ class Bloom(SparseCompressor):
         # (pytorch/deepreduce.py:519-523).
         dense = params.get("dense_tensor", None)
         if dense is not None:
-            positives = ops.bloom_query_positives(packed, m, num_hash, grad_size)
-            new_idxs = _policy_select(positives, num_indices, policy, params, m, num_hash)
+            if policy == "leftmost":
+                # sync-free: output size is host-known (first k positives)
+                new_idxs = ops.bloom_query_leftmost(packed, m, num_hash, grad_size, num_indices)[0]
+            else:
+                positives = ops.bloom_query_positives(packed, m, num_hash, grad_size)
+                new_idxs = _policy_select(positives, num_indices, policy, params, m, num_hash)
             vals = dense.reshape(-1)[new_idxs]
             # side-channel for the wrappers' own-payload cache: decompress of
             # this payload deterministically yields exactly (vals, new_idxs)
@@ -150,8 +154,11 @@ class Bloom(SparseCompressor):
         grad_size = int(torch.Size(shape).numel())
 
         num_hash, m = Bloom._config(num_indices, grad_size, params)
-        positives = ops.bloom_query_positives(packed, m, num_hash, grad_size)
-        idxs = _policy_select(positives, num_indices, policy, params, m, num_hash)
+        if policy == "leftmost":
+            idxs = ops.bloom_query_leftmost(packed, m, num_hash, grad_size, num_indices)[0]
+        else:
+            positives = ops.bloom_query_positives(packed, m, num_hash, grad_size)
+            idxs = _policy_select(positives, num_indices, policy, params, m, num_hash)
         if policy == "p0":
             # every positive gets a value; vals were either FP-aware-read at
             # exactly these positions or truncated/padded to match

@@ -95,6 +95,23 @@ def bloom_query_positives_multi(packed2d: torch.Tensor, m: int, num_hash: int, u
     return torch.cat(outs) if outs else torch.empty(0, dtype=torch.int64), counts
 
 
+def bloom_query_leftmost(packed2d: torch.Tensor, m: int, num_hash: int, universe: int, k: int):
+    """First k positives per rank, int64 [R, k], ascending; NO host sync on
+    the GPU path.  packed2d may be 1-D (single filter).  Requires k <= the
+    number of inserted distinct items (Bloom has no false negatives, so the
+    positive count is always >= that)."""
+    p2 = packed2d if packed2d.dim() == 2 else packed2d.unsqueeze(0)
+    if _want_hip(p2):
+        return _hip.bloom_query_leftmost(p2, m, num_hash, universe, k)
+    rows = []
+    for r in range(p2.shape[0]):
+        pos = _ref.bloom_query_positives(p2[r].contiguous(), m, num_hash, universe)[:k]
+        if pos.numel() < k:
+            pos = torch.nn.functional.pad(pos, (0, k - pos.numel()))
+        rows.append(pos)
+    return torch.stack(rows)
+
+
 def bloom_query_members(packed: torch.Tensor, m: int, num_hash: int, items: torch.Tensor):
     if _want_hip(packed):
         return _hip.bloom_query_members(packed, m, num_hash, items)

@@ -172,6 +172,7 @@ __global__ void bloom_scatter_kernel(const uint64_t* __restrict__ mask, int64_t 
                                      int R, int64_t universe, int64_t chunk,
                                      const int* __restrict__ block_offsets /*[R, nblocks]*/,
                                      const int64_t* __restrict__ rank_base /*[R]*/,
+                                     int64_t max_per_rank /* <=0: unlimited */,
                                      int64_t* __restrict__ out) {
     int64_t start = (int64_t)blockIdx.x * chunk;
     int64_t end = min(start + chunk, universe);
@@ -198,7 +199,8 @@ __global__ void bloom_scatter_kernel(const uint64_t* __restrict__ mask, int64_t 
                 int wbase = 0;
                 for (int w = 0; w < wid; ++w) wbase += wave_cnt[r][w];
                 int prefix = __popcll(ball & ((lane == 63) ? ~0ull >> 1 : ((1ull << lane) - 1)));
-                out[base_s[r] + wbase + prefix] = i;
+                int64_t oi = base_s[r] + wbase + prefix;
+                if (max_per_rank <= 0 || (oi - rank_base[r]) < max_per_rank) out[oi] = i;
             }
         }
         __syncthreads();
@@ -241,9 +243,42 @@ static std::vector<torch::Tensor> query_multi_impl(torch::Tensor bits2d, int64_t
         hipLaunchKernelGGL(bloom_scatter_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
                            (const uint64_t*)mask.data_ptr<int64_t>(), mask_words, R, universe,
                            chunk, offsets.data_ptr<int>(), rank_base_c.data_ptr<int64_t>(),
-                           out.data_ptr<int64_t>());
+                           (int64_t)0, out.data_ptr<int64_t>());
     }
     return {out, rank_totals};
+}
+
+// Sync-free leftmost-k query: returns the FIRST k positives per rank as
+// int64 [R, k] with NO host round-trip (output size is host-known).
+// Invariant: a Bloom filter has no false negatives, so with k <= number of
+// inserted distinct items every row is fully populated; rows are ascending.
+torch::Tensor bloom_query_leftmost(torch::Tensor packed2d, int64_t m, int64_t num_hash,
+                                   int64_t universe, int64_t k_out) {
+    CHECK_CUDA(packed2d);
+    auto bits = packed2d.dim() == 1 ? packed2d.unsqueeze(0).contiguous() : packed2d.contiguous();
+    int R = (int)bits.size(0);
+    TORCH_CHECK(R >= 1 && R <= MAXR, "1..16 ranks supported");
+    int64_t stride_bytes = bits.size(1);
+    int64_t chunk = query_chunk(universe);
+    int64_t nblocks = ceil_div(universe, chunk);
+    auto dev = bits.device();
+    auto counts = torch::empty({R, nblocks}, torch::dtype(torch::kInt32).device(dev));
+    int64_t mask_words = ceil_div(nblocks * chunk, 64);
+    auto mask = torch::empty({R, mask_words}, torch::dtype(torch::kInt64).device(dev));
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(bloom_count_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
+                       bits.data_ptr<uint8_t>(), stride_bytes, R, m, (int)num_hash, universe,
+                       chunk, counts.data_ptr<int>(), (uint64_t*)mask.data_ptr<int64_t>(),
+                       mask_words);
+    auto csum = counts.cumsum(1, torch::kInt32);
+    auto offsets = (csum - counts).to(torch::kInt32).contiguous();
+    auto rank_base = torch::arange(R, torch::dtype(torch::kInt64).device(dev)) * k_out;
+    auto out = torch::zeros({R * k_out}, torch::dtype(torch::kInt64).device(dev));
+    hipLaunchKernelGGL(bloom_scatter_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
+                       (const uint64_t*)mask.data_ptr<int64_t>(), mask_words, R, universe,
+                       chunk, offsets.data_ptr<int>(), rank_base.data_ptr<int64_t>(), k_out,
+                       out.data_ptr<int64_t>());
+    return out.view({R, k_out});
 }
 
 torch::Tensor bloom_query_positives(torch::Tensor packed, int64_t m, int64_t num_hash,
@@ -562,6 +597,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bloom_query_positives", &bloom_query_positives, "Bloom full-universe query (HIP)");
     m.def("bloom_query_positives_multi", &bloom_query_positives_multi,
           "Batched multi-rank Bloom query (HIP): hash once, test R filters");
+    m.def("bloom_query_leftmost", &bloom_query_leftmost,
+          "Sync-free first-k-positives query, [R, k] (HIP)");
     m.def("bloom_query_members", &bloom_query_members, "Bloom membership test (HIP)");
     m.def("qsgd_quantize", &qsgd_quantize, "QSGD quantize (HIP)");
     m.def("qsgd_dequantize", &qsgd_dequantize, "QSGD dequantize (HIP)");

@@ -179,17 +179,11 @@ class IndexCompressor(_WrapperBase):
             return None  # leftmost assumes uniform k (topk sparsifier)
         num_hash, m = Bloom._config(num_indices, numel, self.params)
         bits = torch.stack([p[1].contiguous() for p in payloads])
-        pos, counts = ops.bloom_query_positives_multi(bits, m, num_hash, numel)
-        counts_l = counts.tolist()  # one small sync for slicing
-        base = 0
-        idx_parts, val_parts = [], []
-        for r, c in enumerate(counts_l):
-            n = min(int(c), num_indices)
-            idx_parts.append(pos[base : base + n])
-            val_parts.append(payloads[r][0][:n])
-            base += int(c)
+        # sync-free: [R, k] leftmost positives, one fused scatter-add
+        idxs = ops.bloom_query_leftmost(bits, m, num_hash, numel, num_indices)
+        vals = torch.stack([p[0] for p in payloads])
         dense = torch.zeros(numel, dtype=vals0.dtype, device=vals0.device)
-        dense.index_add_(0, torch.cat(idx_parts), torch.cat(val_parts))
+        dense.index_add_(0, idxs.reshape(-1), vals.reshape(-1))
         return dense.view(shape)
 
 

@@ -146,3 +146,21 @@ def test_bloom_query_multi_parity(hip, dev):
         assert n == singles[r].numel(), f"rank {r}"
         assert torch.equal(pos[base : base + n], singles[r]), f"rank {r}"
         base += n
+
+
+def test_bloom_query_leftmost_parity(hip, dev):
+    """sync-free [R,k] leftmost query == first-k of the full query."""
+    torch.manual_seed(10)
+    universe = 800_000
+    m, k = 250_007, 8
+    filters, expect = [], []
+    kk = 8000
+    for r in range(4):
+        idxs = torch.randperm(universe, device=dev)[:kk]
+        b = hip.bloom_insert(idxs, m, k)
+        filters.append(b)
+        expect.append(hip.bloom_query_positives(b, m, k, universe)[:kk])
+    out = hip.bloom_query_leftmost(torch.stack(filters), m, k, universe, kk)
+    assert out.shape == (4, kk)
+    for r in range(4):
+        assert torch.equal(out[r], expect[r]), f"rank {r}"

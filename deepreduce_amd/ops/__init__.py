@@ -55,8 +55,22 @@ def _want_hip(t: torch.Tensor) -> bool:
 # --------------------------------------------------------------------------
 
 def topk_select(flat: torch.Tensor, k: int):
-    # torch.topk is already a native (rocPRIM-backed) GPU path on ROCm;
-    # a fused HIP select kernel can replace it later if profiling says so.
+    """(vals, idxs) of the k largest-magnitude entries.
+
+    GPU: deterministic two-level radix select (ops/src/hip_ops.hip) — exact
+    top-k up to ties within the top 22 bits of |x| (< 0.012% relative),
+    resolved to the lowest index.  CPU: torch.topk.
+    DEEPREDUCE_TORCH_TOPK=1 forces torch.topk on GPU too.
+    """
+    if (
+        flat.is_cuda
+        and flat.dtype == torch.float32
+        and 1 <= k <= flat.numel()
+        and os.environ.get("DEEPREDUCE_TORCH_TOPK") != "1"
+        and _want_hip(flat)
+    ):
+        out = _hip.topk_select(flat, k)
+        return out[0], out[1]
     return _ref.topk_select(flat, k)
 
 

@@ -484,6 +484,196 @@ torch::Tensor unpack_ints(torch::Tensor stream, int64_t n, int64_t nbits) {
 
 
 // ---------------------------------------------------------------------------
+// Deterministic histogram-threshold top-k select (replaces torch.topk's
+// sort-based path: 3 kernels, no sort, no host sync).
+//
+// |x| as IEEE-754 bits is monotonic for non-negative floats, so top-k by
+// magnitude == top-k by the uint32 key bits(|x|).  An 11-bit histogram of
+// key>>20 (2048 bins) locates the threshold bin; elements in strictly
+// higher bins are all selected, and the tie-bin contributes its LEFTMOST
+// (lowest-index) remainder — fully deterministic.
+// ---------------------------------------------------------------------------
+
+#define TK_BINS 2048
+#define TK_BLOCK 256
+
+// 22-bit monotonic magnitude key: |x| as IEEE bits >> 9.  Two 11-bit radix
+// levels; remaining ties (low 9 mantissa bits, < 0.012% relative) resolve
+// deterministically to the LEFTMOST index.
+__device__ __forceinline__ uint32_t tk_key22(float x) {
+    union { float f; uint32_t u; } c;
+    c.f = fabsf(x);
+    return c.u >> 9;
+}
+
+__global__ void topk_hist_kernel(const float* __restrict__ v, int64_t n,
+                                 int* __restrict__ hist) {
+    __shared__ int lh[TK_BINS];
+    for (int b = threadIdx.x; b < TK_BINS; b += blockDim.x) lh[b] = 0;
+    __syncthreads();
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) atomicAdd(&lh[tk_key22(v[i]) >> 11], 1);
+    __syncthreads();
+    for (int b = threadIdx.x; b < TK_BINS; b += blockDim.x)
+        if (lh[b]) atomicAdd(&hist[b], lh[b]);
+}
+
+// level 2: histogram of the LOW 11 key bits, only for elements whose high
+// bits equal bstar
+__global__ void topk_hist2_kernel(const float* __restrict__ v, int64_t n,
+                                  const int* __restrict__ bstar_p,
+                                  int* __restrict__ hist) {
+    const uint32_t bstar = (uint32_t)*bstar_p;
+    __shared__ int lh[TK_BINS];
+    for (int b = threadIdx.x; b < TK_BINS; b += blockDim.x) lh[b] = 0;
+    __syncthreads();
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        uint32_t key = tk_key22(v[i]);
+        if ((key >> 11) == bstar) atomicAdd(&lh[key & 2047u], 1);
+    }
+    __syncthreads();
+    for (int b = threadIdx.x; b < TK_BINS; b += blockDim.x)
+        if (lh[b]) atomicAdd(&hist[b], lh[b]);
+}
+
+// per-block counts of {key22 > T} and {key22 == T}
+__global__ void topk_count_kernel(const float* __restrict__ v, int64_t n, int64_t chunk,
+                                  const int* __restrict__ thresh_p,
+                                  int* __restrict__ counts /*[2, nblocks]*/) {
+    const uint32_t T = (uint32_t)*thresh_p;
+    int64_t start = (int64_t)blockIdx.x * chunk;
+    int64_t end = min(start + chunk, n);
+    int c0 = 0, c1 = 0;
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+        uint32_t key = tk_key22(v[i]);
+        c0 += (key > T);
+        c1 += (key == T);
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        c0 += __shfl_down(c0, off, WAVE);
+        c1 += __shfl_down(c1, off, WAVE);
+    }
+    __shared__ int w0[TK_BLOCK / WAVE], w1[TK_BLOCK / WAVE];
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) { w0[wid] = c0; w1[wid] = c1; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int t0 = 0, t1 = 0;
+        for (int w = 0; w < TK_BLOCK / WAVE; ++w) { t0 += w0[w]; t1 += w1[w]; }
+        counts[blockIdx.x] = t0;
+        counts[gridDim.x + blockIdx.x] = t1;
+    }
+}
+
+__global__ void topk_scatter_kernel(const float* __restrict__ v, int64_t n, int64_t chunk,
+                                    const int* __restrict__ thresh_p,
+                                    const int* __restrict__ off0 /*[nblocks] excl-cum of c0*/,
+                                    const int* __restrict__ off1 /*[nblocks] excl-cum of c1*/,
+                                    const int* __restrict__ count_above_p,
+                                    int64_t k,
+                                    float* __restrict__ out_v, int64_t* __restrict__ out_i) {
+    const uint32_t T = (uint32_t)*thresh_p;
+    const int64_t above = *count_above_p;   // elements with key22 > T
+    const int64_t need = k - above;         // taken from the tie class
+    int64_t start = (int64_t)blockIdx.x * chunk;
+    int64_t end = min(start + chunk, n);
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    __shared__ int wc0[TK_BLOCK / WAVE], wc1[TK_BLOCK / WAVE];
+    __shared__ int base0_s, base1_s;
+    if (threadIdx.x == 0) { base0_s = off0[blockIdx.x]; base1_s = off1[blockIdx.x]; }
+    __syncthreads();
+    for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        bool in_r = (i < end);
+        uint32_t key = in_r ? tk_key22(v[i]) : 0u;
+        bool p0 = in_r && (key > T);
+        bool p1 = in_r && (key == T);
+        uint64_t b0 = __ballot(p0), b1 = __ballot(p1);
+        if (lane == 0) { wc0[wid] = __popcll(b0); wc1[wid] = __popcll(b1); }
+        __syncthreads();
+        uint64_t below = (lane == 63) ? (~0ull >> 1) : ((1ull << lane) - 1);
+        if (p0) {
+            int wbase = 0;
+            for (int w = 0; w < wid; ++w) wbase += wc0[w];
+            int64_t oi = base0_s + wbase + __popcll(b0 & below);
+            out_i[oi] = i;
+            out_v[oi] = v[i];
+        } else if (p1) {
+            int wbase = 0;
+            for (int w = 0; w < wid; ++w) wbase += wc1[w];
+            int64_t ordinal = base1_s + wbase + __popcll(b1 & below);
+            if (ordinal < need) {
+                out_i[above + ordinal] = i;
+                out_v[above + ordinal] = v[i];
+            }
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int t0 = 0, t1 = 0;
+            for (int w = 0; w < TK_BLOCK / WAVE; ++w) { t0 += wc0[w]; t1 += wc1[w]; }
+            base0_s += t0;
+            base1_s += t1;
+        }
+        __syncthreads();
+    }
+}
+
+std::vector<torch::Tensor> topk_select(torch::Tensor flat, int64_t k) {
+    CHECK_CUDA(flat);
+    auto v = flat.to(torch::kFloat32).contiguous();
+    int64_t n = v.numel();
+    TORCH_CHECK(k >= 1 && k <= n, "k out of range");
+    auto dev = v.device();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+
+    auto hist = torch::zeros({TK_BINS}, torch::dtype(torch::kInt32).device(dev));
+    int hblocks = (int)std::min<int64_t>(ceil_div(n, TK_BLOCK * 16), 2048);
+    hipLaunchKernelGGL(topk_hist_kernel, dim3(hblocks), dim3(TK_BLOCK), 0, stream,
+                       v.data_ptr<float>(), n, hist.data_ptr<int>());
+
+    // level-1 threshold bin (all on device): rev_cum[b] = # of keys >= bin b
+    auto rev_cum = hist.flip(0).cumsum(0, torch::kInt32).flip(0);
+    auto bstar = ((rev_cum >= (int)k).sum(torch::kInt32) - 1).to(torch::kInt32);
+    auto rev_cum_pad = torch::cat({rev_cum, torch::zeros({1}, rev_cum.options())});
+    auto above1 = rev_cum_pad.index({(bstar + 1).to(torch::kLong)}).to(torch::kInt32).reshape({1});
+    auto bstar_c = bstar.reshape({1}).contiguous();
+
+    // level-2 within the tie bin
+    auto hist2 = torch::zeros({TK_BINS}, torch::dtype(torch::kInt32).device(dev));
+    hipLaunchKernelGGL(topk_hist2_kernel, dim3(hblocks), dim3(TK_BLOCK), 0, stream,
+                       v.data_ptr<float>(), n, bstar_c.data_ptr<int>(), hist2.data_ptr<int>());
+    auto rev_cum2 = hist2.flip(0).cumsum(0, torch::kInt32).flip(0);
+    auto k_rem = ((int)k - above1).reshape({});
+    auto bstar2 = ((rev_cum2 >= k_rem).sum(torch::kInt32) - 1).to(torch::kInt32);
+    auto rev_cum2_pad = torch::cat({rev_cum2, torch::zeros({1}, rev_cum2.options())});
+    auto above2 = rev_cum2_pad.index({(bstar2 + 1).to(torch::kLong)}).to(torch::kInt32).reshape({1});
+    auto thresh = (bstar * 2048 + bstar2).to(torch::kInt32).reshape({1}).contiguous();
+    auto count_above = (above1 + above2).to(torch::kInt32).reshape({1}).contiguous();
+
+    int64_t chunk = query_chunk(n);
+    int64_t nblocks = ceil_div(n, chunk);
+    auto counts = torch::empty({2, nblocks}, torch::dtype(torch::kInt32).device(dev));
+    hipLaunchKernelGGL(topk_count_kernel, dim3((int)nblocks), dim3(TK_BLOCK), 0, stream,
+                       v.data_ptr<float>(), n, chunk, thresh.data_ptr<int>(),
+                       counts.data_ptr<int>());
+    auto csum = counts.cumsum(1, torch::kInt32);
+    auto offs = (csum - counts).to(torch::kInt32).contiguous();
+
+    auto out_v = torch::empty({k}, torch::dtype(torch::kFloat32).device(dev));
+    auto out_i = torch::empty({k}, torch::dtype(torch::kInt64).device(dev));
+    auto count_above_c = count_above.contiguous();
+    hipLaunchKernelGGL(topk_scatter_kernel, dim3((int)nblocks), dim3(TK_BLOCK), 0, stream,
+                       v.data_ptr<float>(), n, chunk, thresh.data_ptr<int>(),
+                       offs[0].contiguous().data_ptr<int>(), offs[1].contiguous().data_ptr<int>(),
+                       count_above_c.data_ptr<int>(), k,
+                       out_v.data_ptr<float>(), out_i.data_ptr<int64_t>());
+    return {out_v, out_i};
+}
+
+// ---------------------------------------------------------------------------
 // CPU-native C++ paths (replace the reference's TF C++ CPU ops:
 // bloom_filter_compression.cc / integer_compression.cc) — same wire format
 // and hash math as the HIP kernels, parallelized with at::parallel_for.
@@ -604,6 +794,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("qsgd_dequantize", &qsgd_dequantize, "QSGD dequantize (HIP)");
     m.def("pack_ints", &pack_ints, "n-bit pack (HIP)");
     m.def("unpack_ints", &unpack_ints, "n-bit unpack (HIP)");
+    m.def("topk_select", &topk_select, "histogram-threshold top-k (HIP)");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
     m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");

@@ -164,3 +164,29 @@ def test_bloom_query_leftmost_parity(hip, dev):
     assert out.shape == (4, kk)
     for r in range(4):
         assert torch.equal(out[r], expect[r]), f"rank {r}"
+
+
+def test_topk_select_kernel(hip, dev):
+    """radix-select top-k: exact size/values, boundary within the 22-bit
+    tie band of the true k-th magnitude, deterministic across calls."""
+    from deepreduce_amd.ops import topk_select
+
+    torch.manual_seed(11)
+    for n, k in [(1_000_000, 10_000), (25_000_000, 250_000), (4096, 41), (2048, 2048)]:
+        t = torch.randn(n, device=dev)
+        vals, idxs = topk_select(t, k)
+        assert vals.numel() == k and idxs.numel() == k
+        assert torch.equal(t[idxs], vals)
+        assert idxs.unique().numel() == k  # no duplicates
+        # boundary: min selected |v| >= max unselected |v| within tie band
+        sel_mask = torch.zeros(n, dtype=torch.bool, device=dev)
+        sel_mask[idxs] = True
+        min_sel = vals.abs().min()
+        max_unsel = t.abs()[~sel_mask].max() if k < n else torch.tensor(0.0, device=dev)
+        assert min_sel >= max_unsel * (1 - 2**-12), (min_sel, max_unsel)
+        # determinism
+        v2, i2 = topk_select(t, k)
+        assert torch.equal(i2, idxs)
+        # against torch.topk: selected magnitude sum must match to tie-band
+        tv, _ = torch.topk(t.abs(), k)
+        assert torch.allclose(vals.abs().sum(), tv.sum(), rtol=1e-4)

@@ -95,8 +95,10 @@ def _fit_segments(y: torch.Tensor, segments, degree: int) -> torch.Tensor:
     diag = torch.diagonal(gram, dim1=1, dim2=2)
     ridge = (diag.abs().amax(dim=1, keepdim=True) * 1e-10 + 1e-30)
     gram = gram + torch.diag_embed(ridge.expand(-1, d1))
+    from .. import ops
+
     try:
-        coeffs = torch.linalg.solve(gram, moments.unsqueeze(-1)).squeeze(-1)
+        coeffs = ops.cholesky_solve_small(gram, moments)
     except Exception:  # singular even with ridge: least-squares fallback
         coeffs = torch.linalg.lstsq(gram, moments.unsqueeze(-1)).solution.squeeze(-1)
     return coeffs  # [S, d1]

@@ -150,6 +150,14 @@ def unpack_ints(stream: torch.Tensor, n: int, nbits: int) -> torch.Tensor:
     return _ref.unpack_ints(stream, n, nbits)
 
 
+def cholesky_solve_small(G: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Batched SPD solve for [S, d<=8, d] systems (polyfit normal equations).
+    GPU: one-thread-per-system in-register Cholesky; CPU: torch."""
+    if _want_hip(G):
+        return _hip.cholesky_solve_small(G, b)
+    return torch.linalg.solve(G, b.unsqueeze(-1)).squeeze(-1)
+
+
 def qsgd_quantize(vals: torch.Tensor, quantum_num: int, bucket_size: int):
     if _want_hip(vals):
         return _hip.qsgd_quantize(vals, quantum_num, bucket_size)

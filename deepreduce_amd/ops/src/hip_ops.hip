@@ -674,6 +674,70 @@ std::vector<torch::Tensor> topk_select(torch::Tensor flat, int64_t k) {
 }
 
 // ---------------------------------------------------------------------------
+// Batched tiny SPD solve (polyfit normal equations): one thread per system,
+// in-register Cholesky for d1 <= 8.  Replaces the reference's per-segment
+// CPU inverse round-trip (pytorch/deepreduce.py:331-334) and rocSOLVER's
+// small-batch LU (milliseconds per call) with a microsecond kernel.
+// ---------------------------------------------------------------------------
+
+#define SOLVE_MAXD 8
+
+__global__ void cholesky_solve_kernel(const double* __restrict__ G /*[S,d,d]*/,
+                                      const double* __restrict__ b /*[S,d]*/,
+                                      int S, int d, double* __restrict__ x /*[S,d]*/) {
+    int s = blockIdx.x * blockDim.x + threadIdx.x;
+    if (s >= S) return;
+    double L[SOLVE_MAXD][SOLVE_MAXD];
+    double y[SOLVE_MAXD];
+    const double* g = G + (int64_t)s * d * d;
+    const double* bb = b + (int64_t)s * d;
+    // Cholesky G = L L^T (lower)
+    for (int i = 0; i < d; ++i) {
+        for (int j = 0; j <= i; ++j) {
+            double sum = g[i * d + j];
+            for (int p = 0; p < j; ++p) sum -= L[i][p] * L[j][p];
+            if (i == j) {
+                L[i][j] = sqrt(sum > 1e-300 ? sum : 1e-300);
+            } else {
+                L[i][j] = sum / L[j][j];
+            }
+        }
+    }
+    // forward: L y = b
+    for (int i = 0; i < d; ++i) {
+        double sum = bb[i];
+        for (int p = 0; p < i; ++p) sum -= L[i][p] * y[p];
+        y[i] = sum / L[i][i];
+    }
+    // backward: L^T x = y
+    double* xx = x + (int64_t)s * d;
+    for (int i = d - 1; i >= 0; --i) {
+        double sum = y[i];
+        for (int p = i + 1; p < d; ++p) sum -= L[p][i] * xx[p];
+        xx[i] = sum / L[i][i];
+    }
+}
+
+torch::Tensor cholesky_solve_small(torch::Tensor G, torch::Tensor b) {
+    CHECK_CUDA(G);
+    auto g = G.to(torch::kFloat64).contiguous();
+    auto bb = b.to(torch::kFloat64).contiguous();
+    int S = (int)g.size(0);
+    int d = (int)g.size(1);
+    TORCH_CHECK(d <= SOLVE_MAXD, "d <= 8");
+    auto x = torch::empty({S, d}, g.options());
+    if (S > 0) {
+        int threads = 64;
+        int blocks = (S + threads - 1) / threads;
+        hipStream_t stream = at::hip::getCurrentHIPStream();
+        hipLaunchKernelGGL(cholesky_solve_kernel, dim3(blocks), dim3(threads), 0, stream,
+                           g.data_ptr<double>(), bb.data_ptr<double>(), S, d,
+                           x.data_ptr<double>());
+    }
+    return x;
+}
+
+// ---------------------------------------------------------------------------
 // CPU-native C++ paths (replace the reference's TF C++ CPU ops:
 // bloom_filter_compression.cc / integer_compression.cc) — same wire format
 // and hash math as the HIP kernels, parallelized with at::parallel_for.
@@ -795,6 +859,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pack_ints", &pack_ints, "n-bit pack (HIP)");
     m.def("unpack_ints", &unpack_ints, "n-bit unpack (HIP)");
     m.def("topk_select", &topk_select, "histogram-threshold top-k (HIP)");
+    m.def("cholesky_solve_small", &cholesky_solve_small, "batched tiny SPD solve (HIP)");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
     m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");

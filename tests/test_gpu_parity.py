@@ -190,3 +190,14 @@ def test_topk_select_kernel(hip, dev):
         # against torch.topk: selected magnitude sum must match to tie-band
         tv, _ = torch.topk(t.abs(), k)
         assert torch.allclose(vals.abs().sum(), tv.sum(), rtol=1e-4)
+
+
+def test_cholesky_solve_small_parity(hip, dev):
+    torch.manual_seed(12)
+    S, d = 37, 6
+    A = torch.randn(S, d, d, dtype=torch.float64, device=dev)
+    G = A @ A.transpose(1, 2) + torch.eye(d, dtype=torch.float64, device=dev) * 1e-3
+    b = torch.randn(S, d, dtype=torch.float64, device=dev)
+    x = hip.cholesky_solve_small(G, b)
+    x_ref = torch.linalg.solve(G.cpu(), b.cpu().unsqueeze(-1)).squeeze(-1)
+    assert torch.allclose(x.cpu(), x_ref, rtol=1e-8, atol=1e-10)

@@ -56,7 +56,13 @@ class ResidualMemory(Memory):
             decompressed = own(tensor_compressed, ctx, name)
         else:
             decompressed = compressor.decompress(tensor_compressed, ctx)
-        self.residuals[name] = tensor - decompressed
+        r = self.residuals.get(name)
+        if r is not None and r.shape == tensor.shape:
+            # in-place into the persistent buffer: keeps the residual's
+            # storage stable so the whole exchange is hipGraph-capturable
+            torch.sub(tensor, decompressed.view_as(tensor), out=r)
+        else:
+            self.residuals[name] = tensor - decompressed.view_as(tensor)
 
     # Checkpoint support (absent in the reference — residuals were lost on
     # restart, tensorflow/deepreduce.py:39; we do better).

@@ -83,17 +83,20 @@ torch::Tensor bloom_insert(torch::Tensor idxs, int64_t m, int64_t num_hash) {
     CHECK_CUDA(idxs);
     auto items = idxs.to(torch::kInt64).contiguous();
     int64_t nwords = ceil_div(m, 32);
-    auto words = torch::zeros({nwords}, torch::dtype(torch::kInt32).device(idxs.device()));
+    auto words = torch::empty({nwords}, torch::dtype(torch::kInt32).device(idxs.device()));
     int64_t n = items.numel();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    (void)hipMemsetAsync(words.data_ptr<int32_t>(), 0, (size_t)nwords * 4, stream);
     if (n > 0) {
         int threads = 256;
         int blocks = (int)std::min<int64_t>(ceil_div(n, threads), 4096);
-        hipStream_t stream = at::hip::getCurrentHIPStream();
         hipLaunchKernelGGL(bloom_insert_kernel, dim3(blocks), dim3(threads), 0, stream,
                            items.data_ptr<int64_t>(), n, (int)num_hash, m,
                            (uint32_t*)words.data_ptr<int32_t>());
     }
-    return words.view(torch::kUInt8).narrow(0, 0, ceil_div(m, 8)).contiguous();
+    // narrow at offset 0 of a contiguous 1-D tensor is itself contiguous: a
+    // VIEW, not a copy
+    return words.view(torch::kUInt8).narrow(0, 0, ceil_div(m, 8));
 }
 
 // ---------------------------------------------------------------------------
@@ -167,11 +170,54 @@ __global__ void bloom_count_kernel(const uint8_t* __restrict__ bits, int64_t str
     }
 }
 
+// Exclusive scan along dim 1 of an int32 [R, n] matrix, one block per row.
+// Replaces the cumsum/sub/to/contiguous torch-op chain in the query and
+// top-k drivers (the launch-bound hot path: SURVEY.md sect. 3.1 note).
+__global__ void exclusive_scan_rows_kernel(const int* __restrict__ in, int64_t n,
+                                           int* __restrict__ out,
+                                           int* __restrict__ row_totals /*nullable [R]*/) {
+    const int64_t row = blockIdx.x;
+    const int* src = in + row * n;
+    int* dst = out + row * n;
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    __shared__ int wave_tot[QBLOCK / WAVE];
+    __shared__ int carry;
+    if (threadIdx.x == 0) carry = 0;
+    __syncthreads();
+    for (int64_t i0 = 0; i0 < n; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        int v = (i < n) ? src[i] : 0;
+        // inclusive wave scan
+        int incl = v;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        if (lane == WAVE - 1) wave_tot[wid] = incl;
+        __syncthreads();
+        int wbase = 0;
+        for (int w = 0; w < wid; ++w) wbase += wave_tot[w];
+        if (i < n) dst[i] = carry + wbase + incl - v;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int t = 0;
+            for (int w = 0; w < QBLOCK / WAVE; ++w) t += wave_tot[w];
+            carry += t;
+        }
+        __syncthreads();
+    }
+    if (row_totals != nullptr && threadIdx.x == 0) row_totals[row] = carry;
+}
+
 // Pass 2: ordered compaction of the predicate bit-plane (no hashing).
+// rank_base: per-rank output base — explicit device array, or r*rank_stride
+// when rank_base == nullptr (keeps the sync-free leftmost path free of
+// torch-op glue).
 __global__ void bloom_scatter_kernel(const uint64_t* __restrict__ mask, int64_t mask_stride,
                                      int R, int64_t universe, int64_t chunk,
                                      const int* __restrict__ block_offsets /*[R, nblocks]*/,
-                                     const int64_t* __restrict__ rank_base /*[R]*/,
+                                     const int64_t* __restrict__ rank_base /*[R] or null*/,
+                                     int64_t rank_stride,
                                      int64_t max_per_rank /* <=0: unlimited */,
                                      int64_t* __restrict__ out) {
     int64_t start = (int64_t)blockIdx.x * chunk;
@@ -179,10 +225,13 @@ __global__ void bloom_scatter_kernel(const uint64_t* __restrict__ mask, int64_t 
     int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
     __shared__ int wave_cnt[MAXR][QBLOCK / WAVE];
     __shared__ int base_s[MAXR];
-    if (threadIdx.x < MAXR && threadIdx.x < R)
+    __shared__ int64_t rbase_s[MAXR];
+    if (threadIdx.x < MAXR && threadIdx.x < R) {
+        rbase_s[threadIdx.x] =
+            rank_base ? rank_base[threadIdx.x] : (int64_t)threadIdx.x * rank_stride;
         base_s[threadIdx.x] =
-            block_offsets[threadIdx.x * gridDim.x + blockIdx.x] +
-            (int)rank_base[threadIdx.x];
+            block_offsets[threadIdx.x * gridDim.x + blockIdx.x] + (int)rbase_s[threadIdx.x];
+    }
     __syncthreads();
     for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
         int64_t i = i0 + threadIdx.x;
@@ -200,7 +249,7 @@ __global__ void bloom_scatter_kernel(const uint64_t* __restrict__ mask, int64_t 
                 for (int w = 0; w < wid; ++w) wbase += wave_cnt[r][w];
                 int prefix = __popcll(ball & ((lane == 63) ? ~0ull >> 1 : ((1ull << lane) - 1)));
                 int64_t oi = base_s[r] + wbase + prefix;
-                if (max_per_rank <= 0 || (oi - rank_base[r]) < max_per_rank) out[oi] = i;
+                if (max_per_rank <= 0 || (oi - rbase_s[r]) < max_per_rank) out[oi] = i;
             }
         }
         __syncthreads();
@@ -232,18 +281,21 @@ static std::vector<torch::Tensor> query_multi_impl(torch::Tensor bits2d, int64_t
                        bits.data_ptr<uint8_t>(), stride_bytes, R, m, (int)num_hash, universe,
                        chunk, counts.data_ptr<int>(), (uint64_t*)mask.data_ptr<int64_t>(),
                        mask_words);
-    auto csum = counts.cumsum(1, torch::kInt32);
-    auto offsets = (csum - counts).to(torch::kInt32).contiguous();
-    auto rank_totals = csum.select(1, nblocks - 1).to(torch::kInt64);
+    auto offsets = torch::empty({R, nblocks}, torch::dtype(torch::kInt32).device(dev));
+    auto row_totals = torch::empty({R}, torch::dtype(torch::kInt32).device(dev));
+    hipLaunchKernelGGL(exclusive_scan_rows_kernel, dim3(R), dim3(QBLOCK), 0, stream,
+                       counts.data_ptr<int>(), nblocks, offsets.data_ptr<int>(),
+                       row_totals.data_ptr<int>());
+    auto rank_totals = row_totals.to(torch::kInt64);
     auto rank_base = rank_totals.cumsum(0) - rank_totals;
-    int64_t total = (int64_t)rank_totals.sum().item<int64_t>();  // one sync
+    int64_t total = (int64_t)rank_totals.sum().item<int64_t>();  // one sync (ragged path)
     auto out = torch::empty({total}, torch::dtype(torch::kInt64).device(dev));
     if (total > 0) {
         auto rank_base_c = rank_base.contiguous();
         hipLaunchKernelGGL(bloom_scatter_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
                            (const uint64_t*)mask.data_ptr<int64_t>(), mask_words, R, universe,
                            chunk, offsets.data_ptr<int>(), rank_base_c.data_ptr<int64_t>(),
-                           (int64_t)0, out.data_ptr<int64_t>());
+                           (int64_t)0, (int64_t)0, out.data_ptr<int64_t>());
     }
     return {out, rank_totals};
 }
@@ -262,21 +314,24 @@ torch::Tensor bloom_query_leftmost(torch::Tensor packed2d, int64_t m, int64_t nu
     int64_t chunk = query_chunk(universe);
     int64_t nblocks = ceil_div(universe, chunk);
     auto dev = bits.device();
-    auto counts = torch::empty({R, nblocks}, torch::dtype(torch::kInt32).device(dev));
+    // one int32 workspace: counts [R, nblocks] | offsets [R, nblocks]
+    auto ws = torch::empty({2 * R * nblocks}, torch::dtype(torch::kInt32).device(dev));
+    int* counts = ws.data_ptr<int>();
+    int* offsets = counts + R * nblocks;
     int64_t mask_words = ceil_div(nblocks * chunk, 64);
     auto mask = torch::empty({R, mask_words}, torch::dtype(torch::kInt64).device(dev));
     hipStream_t stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(bloom_count_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
                        bits.data_ptr<uint8_t>(), stride_bytes, R, m, (int)num_hash, universe,
-                       chunk, counts.data_ptr<int>(), (uint64_t*)mask.data_ptr<int64_t>(),
+                       chunk, counts, (uint64_t*)mask.data_ptr<int64_t>(),
                        mask_words);
-    auto csum = counts.cumsum(1, torch::kInt32);
-    auto offsets = (csum - counts).to(torch::kInt32).contiguous();
-    auto rank_base = torch::arange(R, torch::dtype(torch::kInt64).device(dev)) * k_out;
-    auto out = torch::zeros({R * k_out}, torch::dtype(torch::kInt64).device(dev));
+    hipLaunchKernelGGL(exclusive_scan_rows_kernel, dim3(R), dim3(QBLOCK), 0, stream,
+                       counts, nblocks, offsets, (int*)nullptr);
+    auto out = torch::empty({R * k_out}, torch::dtype(torch::kInt64).device(dev));
+    (void)hipMemsetAsync(out.data_ptr<int64_t>(), 0, (size_t)(R * k_out) * 8, stream);
     hipLaunchKernelGGL(bloom_scatter_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
                        (const uint64_t*)mask.data_ptr<int64_t>(), mask_words, R, universe,
-                       chunk, offsets.data_ptr<int>(), rank_base.data_ptr<int64_t>(), k_out,
+                       chunk, offsets, (const int64_t*)nullptr, k_out, k_out,
                        out.data_ptr<int64_t>());
     return out.view({R, k_out});
 }
@@ -539,6 +594,65 @@ __global__ void topk_hist2_kernel(const float* __restrict__ v, int64_t n,
         if (lh[b]) atomicAdd(&hist[b], lh[b]);
 }
 
+// Derive the threshold bin from a 2048-bin histogram on-device (one
+// wavefront): bstar = largest bin b with suffix_sum(b) >= target, above =
+// suffix_sum(bstar+1).  Replaces the flip/cumsum/cat/index torch-op chain
+// that made the old driver launch-bound (~20 composite ops per tensor).
+// scalars layout: [0]=bstar1 [1]=above1 [2]=thresh22 [3]=count_above.
+// level 1: target = k.  level 2: target = k - above1; also finalizes
+// thresh22 = bstar1*2048 + bstar2 and count_above = above1 + above2.
+__global__ void topk_thresh_kernel(const int* __restrict__ hist, int64_t k, int level,
+                                   int* __restrict__ sc) {
+    const int lane = threadIdx.x;               // single wave of 64
+    const int target = (level == 1) ? (int)k : (int)k - sc[1];
+    const int SEG = TK_BINS / WAVE;             // 32 bins per lane
+    int h[SEG];
+    int seg_sum = 0;
+    for (int j = 0; j < SEG; ++j) {
+        h[j] = hist[lane * SEG + j];
+        seg_sum += h[j];
+    }
+    // suffix over lanes: sum of seg_sum for lanes strictly greater
+    int suffix_excl = 0;
+    {
+        int acc = seg_sum;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int other = __shfl_down(acc, off, WAVE);
+            if (lane + off < WAVE) acc += other;
+        }
+        // acc now = suffix inclusive at this lane; recover exclusive
+        suffix_excl = acc - seg_sum;
+    }
+    // walk own segment from the right: first j (largest) with suffix >= target
+    int found_b = -1, found_above = 0;
+    int acc = 0;
+    for (int j = SEG - 1; j >= 0; --j) {
+        int suffix_here = suffix_excl + acc + h[j];
+        if (suffix_here >= target) {
+            found_b = lane * SEG + j;
+            found_above = suffix_excl + acc;  // suffix(b+1)
+            break;
+        }
+        acc += h[j];
+    }
+    // max-reduce found_b across the wave; winning lane writes results
+    int best = found_b;
+    for (int off = 1; off < WAVE; off <<= 1) {
+        int other = __shfl_down(best, off, WAVE);
+        if (lane + off < WAVE && other > best) best = other;
+    }
+    best = __shfl(best, 0, WAVE);
+    if (found_b == best && found_b >= 0) {
+        if (level == 1) {
+            sc[0] = found_b;
+            sc[1] = found_above;
+        } else {
+            sc[2] = sc[0] * TK_BINS + found_b;
+            sc[3] = sc[1] + found_above;
+        }
+    }
+}
+
 // per-block counts of {key22 > T} and {key22 == T}
 __global__ void topk_count_kernel(const float* __restrict__ v, int64_t n, int64_t chunk,
                                   const int* __restrict__ thresh_p,
@@ -629,46 +743,38 @@ std::vector<torch::Tensor> topk_select(torch::Tensor flat, int64_t k) {
     auto dev = v.device();
     hipStream_t stream = at::hip::getCurrentHIPStream();
 
-    auto hist = torch::zeros({TK_BINS}, torch::dtype(torch::kInt32).device(dev));
-    int hblocks = (int)std::min<int64_t>(ceil_div(n, TK_BLOCK * 16), 2048);
-    hipLaunchKernelGGL(topk_hist_kernel, dim3(hblocks), dim3(TK_BLOCK), 0, stream,
-                       v.data_ptr<float>(), n, hist.data_ptr<int>());
-
-    // level-1 threshold bin (all on device): rev_cum[b] = # of keys >= bin b
-    auto rev_cum = hist.flip(0).cumsum(0, torch::kInt32).flip(0);
-    auto bstar = ((rev_cum >= (int)k).sum(torch::kInt32) - 1).to(torch::kInt32);
-    auto rev_cum_pad = torch::cat({rev_cum, torch::zeros({1}, rev_cum.options())});
-    auto above1 = rev_cum_pad.index({(bstar + 1).to(torch::kLong)}).to(torch::kInt32).reshape({1});
-    auto bstar_c = bstar.reshape({1}).contiguous();
-
-    // level-2 within the tie bin
-    auto hist2 = torch::zeros({TK_BINS}, torch::dtype(torch::kInt32).device(dev));
-    hipLaunchKernelGGL(topk_hist2_kernel, dim3(hblocks), dim3(TK_BLOCK), 0, stream,
-                       v.data_ptr<float>(), n, bstar_c.data_ptr<int>(), hist2.data_ptr<int>());
-    auto rev_cum2 = hist2.flip(0).cumsum(0, torch::kInt32).flip(0);
-    auto k_rem = ((int)k - above1).reshape({});
-    auto bstar2 = ((rev_cum2 >= k_rem).sum(torch::kInt32) - 1).to(torch::kInt32);
-    auto rev_cum2_pad = torch::cat({rev_cum2, torch::zeros({1}, rev_cum2.options())});
-    auto above2 = rev_cum2_pad.index({(bstar2 + 1).to(torch::kLong)}).to(torch::kInt32).reshape({1});
-    auto thresh = (bstar * 2048 + bstar2).to(torch::kInt32).reshape({1}).contiguous();
-    auto count_above = (above1 + above2).to(torch::kInt32).reshape({1}).contiguous();
-
+    // Single int32 workspace; 7 kernels, zero composite torch ops, zero
+    // host syncs.  Layout: hist1[2048] | hist2[2048] | scalars[4] |
+    // counts[2*nblocks] | offsets[2*nblocks].
     int64_t chunk = query_chunk(n);
     int64_t nblocks = ceil_div(n, chunk);
-    auto counts = torch::empty({2, nblocks}, torch::dtype(torch::kInt32).device(dev));
+    auto ws = torch::empty({2 * TK_BINS + 4 + 4 * nblocks},
+                           torch::dtype(torch::kInt32).device(dev));
+    int* hist1 = ws.data_ptr<int>();
+    int* hist2 = hist1 + TK_BINS;
+    int* sc = hist2 + TK_BINS;
+    int* counts = sc + 4;
+    int* offs = counts + 2 * nblocks;
+    (void)hipMemsetAsync(hist1, 0, (size_t)(2 * TK_BINS + 4) * sizeof(int), stream);
+
+    int hblocks = (int)std::min<int64_t>(ceil_div(n, TK_BLOCK * 16), 2048);
+    hipLaunchKernelGGL(topk_hist_kernel, dim3(hblocks), dim3(TK_BLOCK), 0, stream,
+                       v.data_ptr<float>(), n, hist1);
+    hipLaunchKernelGGL(topk_thresh_kernel, dim3(1), dim3(WAVE), 0, stream, hist1, k, 1, sc);
+    hipLaunchKernelGGL(topk_hist2_kernel, dim3(hblocks), dim3(TK_BLOCK), 0, stream,
+                       v.data_ptr<float>(), n, &sc[0], hist2);
+    hipLaunchKernelGGL(topk_thresh_kernel, dim3(1), dim3(WAVE), 0, stream, hist2, k, 2, sc);
+
     hipLaunchKernelGGL(topk_count_kernel, dim3((int)nblocks), dim3(TK_BLOCK), 0, stream,
-                       v.data_ptr<float>(), n, chunk, thresh.data_ptr<int>(),
-                       counts.data_ptr<int>());
-    auto csum = counts.cumsum(1, torch::kInt32);
-    auto offs = (csum - counts).to(torch::kInt32).contiguous();
+                       v.data_ptr<float>(), n, chunk, &sc[2], counts);
+    hipLaunchKernelGGL(exclusive_scan_rows_kernel, dim3(2), dim3(QBLOCK), 0, stream,
+                       counts, nblocks, offs, (int*)nullptr);
 
     auto out_v = torch::empty({k}, torch::dtype(torch::kFloat32).device(dev));
     auto out_i = torch::empty({k}, torch::dtype(torch::kInt64).device(dev));
-    auto count_above_c = count_above.contiguous();
     hipLaunchKernelGGL(topk_scatter_kernel, dim3((int)nblocks), dim3(TK_BLOCK), 0, stream,
-                       v.data_ptr<float>(), n, chunk, thresh.data_ptr<int>(),
-                       offs[0].contiguous().data_ptr<int>(), offs[1].contiguous().data_ptr<int>(),
-                       count_above_c.data_ptr<int>(), k,
+                       v.data_ptr<float>(), n, chunk, &sc[2],
+                       offs, offs + nblocks, &sc[3], k,
                        out_v.data_ptr<float>(), out_i.data_ptr<int64_t>());
     return {out_v, out_i};
 }

@@ -1,18 +1,27 @@
 """Distributed optimizer / gradient-hook layer (GRACE DistributedOptimizer
 equivalent — the part the reference leaves to grace_dl, SURVEY.md sect. 2.5).
 
-Two modes:
+Modes:
   * `reduce_gradients(model, grc)` — synchronous: call between backward and
     optimizer.step().
   * `DistributedOptimizer(opt, grc, model)` — wraps an optimizer; gradients
-    are compressed+exchanged in .step() before the inner step.  On CUDA the
-    exchange runs per-parameter in reverse registration order so decompress
-    of early buckets overlaps compression of later ones on the comm stream.
+    are compressed+exchanged in .step() before the inner step.
+  * hipGraph capture (MI355X-first): the leftmost-bloom/topk/qsgd pipeline
+    is sync-free by construction (no .item()/.cpu() anywhere, payload sizes
+    host-known), so the ENTIRE per-step exchange — compensate, top-k
+    select, bloom insert/query, payload fusion, collective, fused
+    multi-rank decompress, residual update — is captured once into a
+    hipGraph and replayed as a single launch.  This removes the per-tensor
+    launch storm (~10k dispatches/step measured eager on ResNet-50,
+    profiles/) that dominates step time.  Enabled automatically when the
+    pipeline is graph-safe; DEEPREDUCE_GRAPH=0 disables.
 
 Gradients are compressed in float32 regardless of compute dtype (the wire
 volume accounting and all codecs are float32, matching the reference).
 """
 from __future__ import annotations
+
+import os
 
 import torch
 
@@ -34,20 +43,92 @@ def reduce_gradients(model: torch.nn.Module, grc, names=None):
     return total_bytes
 
 
-class DistributedOptimizer:
-    """Optimizer wrapper: exchange compressed grads, then inner step."""
+def _graph_safe(grc) -> bool:
+    """True when the configured pipeline has no host syncs or data-dependent
+    payload sizes: leftmost-policy bloom (sync-free query), qsgd, plain
+    topk/randomk/none — NOT polyfit (segment count depends on num_pos),
+    p0/conflict_sets/random policies, or threshold sparsification."""
+    if os.environ.get("DEEPREDUCE_GRAPH") == "0":
+        return False
+    comp = getattr(grc, "compressor", None)
+    params = getattr(comp, "params", None)
+    if params is None:  # bare sparsifier
+        from .compressors import ThresholdCompressor
 
-    def __init__(self, optimizer: torch.optim.Optimizer, grc, model: torch.nn.Module):
+        return not isinstance(comp, ThresholdCompressor)
+    mode = params.get("deepreduce")
+    if not mode:
+        return True
+    value = params.get("value", "polyfit")
+    index = params.get("index", "bloom")
+    policy = params.get("policy", "leftmost")
+    if mode in ("index", "both") and (index != "bloom" or policy != "leftmost"):
+        return False
+    if mode in ("value", "both") and value not in ("qsgd",):
+        return False
+    if params.get("micro-benchmark"):
+        return False  # timing prints sync
+    return True
+
+
+class DistributedOptimizer:
+    """Optimizer wrapper: exchange compressed grads, then inner step.
+
+    On ROCm, the exchange is hipGraph-captured after `graph_warmup` eager
+    steps when the pipeline is graph-safe (see _graph_safe); capture
+    failure falls back to eager permanently.
+    """
+
+    def __init__(self, optimizer: torch.optim.Optimizer, grc, model: torch.nn.Module,
+                 use_graph: bool | None = None, graph_warmup: int = 3):
         self.optimizer = optimizer
         self.grc = grc
         self.model = model
         self.last_wire_bytes = 0
+        self._graph = None
+        self._graph_calls = 0
+        self._graph_warmup = max(2, graph_warmup)  # residuals must exist
+        self._use_graph = use_graph if use_graph is not None else _graph_safe(grc)
+        self._graph_wire_bytes = 0
 
     def zero_grad(self, set_to_none: bool = True):
+        if not set_to_none:
+            grads = [p.grad for p in self.model.parameters() if p.grad is not None]
+            if grads:
+                torch._foreach_zero_(grads)  # one fused launch, not one per tensor
+                return
         self.optimizer.zero_grad(set_to_none=set_to_none)
 
+    def _exchange(self):
+        if not (self._use_graph and torch.cuda.is_available()
+                and next(self.model.parameters()).is_cuda):
+            self.last_wire_bytes = reduce_gradients(self.model, self.grc)
+            return
+        if self._graph is not None:
+            self._graph.replay()
+            self.last_wire_bytes = self._graph_wire_bytes
+            return
+        self._graph_calls += 1
+        if self._graph_calls <= self._graph_warmup:
+            self.last_wire_bytes = reduce_gradients(self.model, self.grc)
+            return
+        # Capture.  Storage stability: p.grad buffers (zero_grad
+        # set_to_none=False keeps them), residual buffers (ResidualMemory
+        # updates in place), and graph-pool intermediates.
+        try:
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._graph_wire_bytes = reduce_gradients(self.model, self.grc)
+            self._graph = g
+            self.last_wire_bytes = self._graph_wire_bytes
+        except Exception:
+            self._use_graph = False
+            torch.cuda.synchronize()
+            self.last_wire_bytes = reduce_gradients(self.model, self.grc)
+
     def step(self, closure=None):
-        self.last_wire_bytes = reduce_gradients(self.model, self.grc)
+        self._exchange()
         return self.optimizer.step(closure)
 
     @property

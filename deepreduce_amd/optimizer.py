@@ -48,7 +48,10 @@ def _graph_safe(grc) -> bool:
     payload sizes: leftmost-policy bloom (sync-free query), qsgd, plain
     topk/randomk/none — NOT polyfit (segment count depends on num_pos),
     p0/conflict_sets/random policies, or threshold sparsification."""
-    if os.environ.get("DEEPREDUCE_GRAPH") == "0":
+    # Opt-in while hipGraph replay is being qualified on real hardware;
+    # flipped to default-on once the soak run below is green (see
+    # profiles/NOTES.md).  DEEPREDUCE_GRAPH=0 always disables.
+    if os.environ.get("DEEPREDUCE_GRAPH", "0") != "1":
         return False
     comp = getattr(grc, "compressor", None)
     params = getattr(comp, "params", None)
@@ -114,8 +117,14 @@ class DistributedOptimizer:
             return
         # Capture.  Storage stability: p.grad buffers (zero_grad
         # set_to_none=False keeps them), residual buffers (ResidualMemory
-        # updates in place), and graph-pool intermediates.
+        # updates in place), and graph-pool intermediates.  Canonical
+        # recipe: warm the exact capture path up on a side stream first.
         try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                reduce_gradients(self.model, self.grc)
+            torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
@@ -124,6 +133,7 @@ class DistributedOptimizer:
             self.last_wire_bytes = self._graph_wire_bytes
         except Exception:
             self._use_graph = False
+            self._graph = None
             torch.cuda.synchronize()
             self.last_wire_bytes = reduce_gradients(self.model, self.grc)
 

@@ -24,6 +24,13 @@ class Memory:
     def update(self, tensor, name, compressor, tensor_compressed, ctx):
         pass
 
+    # batched variants used by the fused (step_many) exchange path
+    def compensate_many(self, tensors, names):
+        return [self.compensate(t, n) for t, n in zip(tensors, names)]
+
+    def update_many(self, tensors, names, decompressed):
+        pass
+
     def state_dict(self):
         return {}
 
@@ -63,6 +70,30 @@ class ResidualMemory(Memory):
             torch.sub(tensor, decompressed.view_as(tensor), out=r)
         else:
             self.residuals[name] = tensor - decompressed.view_as(tensor)
+
+    def compensate_many(self, tensors, names):
+        """Batched compensate: residual buffers are materialized (zeros) on
+        first sight so the math is a single _foreach op afterwards; buffers
+        inherit the gradient's layout (zeros_like) so channels-last convs
+        stay on the TensorIterator fast path."""
+        rs = []
+        for t, n in zip(tensors, names):
+            r = self.residuals.get(n)
+            if r is None or r.shape != t.shape:
+                r = torch.zeros_like(t)
+                self.residuals[n] = r
+            rs.append(r)
+        if self.beta == 1.0 and self.gamma == 1.0:
+            return torch._foreach_add(rs, tensors)
+        out = torch._foreach_mul(tensors, self.gamma)
+        torch._foreach_add_(out, rs, alpha=self.beta)
+        return out
+
+    def update_many(self, tensors, names, decompressed):
+        """residual <- compensated - decompressed, in place (graph-stable)."""
+        rs = [self.residuals[n] for n in names]
+        torch._foreach_copy_(rs, list(tensors))
+        torch._foreach_sub_(rs, [d.view_as(t) for d, t in zip(decompressed, tensors)])
 
     # Checkpoint support (absent in the reference — residuals were lost on
     # restart, tensorflow/deepreduce.py:39; we do better).

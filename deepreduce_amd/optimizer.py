@@ -28,14 +28,24 @@ import torch
 __all__ = ["reduce_gradients", "DistributedOptimizer"]
 
 
-def reduce_gradients(model: torch.nn.Module, grc, names=None):
-    """Compress+exchange every .grad in place.  Returns total wire bytes."""
+def reduce_gradients(model: torch.nn.Module, grc, names=None, fused: bool = True):
+    """Compress+exchange every .grad in place.  Returns total wire bytes.
+
+    `fused` (default) routes through the communicator's step_many — the
+    whole-model bucket-fused exchange (one collective per step instead of
+    one per tensor).  Set False for the reference-style per-tensor loop.
+    """
+    params = [(n, p) for n, p in reversed(list(model.named_parameters()))
+              if p.grad is not None]
+    if fused and hasattr(grc, "step_many") and os.environ.get("DEEPREDUCE_FUSED", "1") == "1":
+        named = [(n, p.grad.data.float()) for n, p in params]
+        reduced = grc.step_many(named)
+        for (name, p), r in zip(params, reduced):
+            p.grad.data.copy_(r.view_as(p.grad.data))
+        return getattr(grc, "last_wire_bytes", 0)
     total_bytes = 0
-    params = list(model.named_parameters())
     # reverse order: last layers' grads are ready first after backward
-    for name, p in reversed(params):
-        if p.grad is None:
-            continue
+    for name, p in params:
         g32 = p.grad.data.float()
         reduced = grc.step(g32, name)
         p.grad.data.copy_(reduced.view_as(p.grad.data))

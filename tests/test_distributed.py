@@ -165,3 +165,57 @@ def test_ragged_allgather_threshold():
         p.join(timeout=60)
     for rank, same in results:
         assert same is True, f"rank {rank}: {same}"
+
+
+def _run_fused_equivalence(rank, world, mode, q):
+    try:
+        _init(rank, world)
+        from deepreduce_amd import deepreduce_from_params
+
+        params = {
+            "compressor": "topk",
+            "memory": "residual",
+            "communicator": "allgather",
+            "compress_ratio": 0.01,
+        }
+        if mode == "index":
+            params.update({"deepreduce": "index", "index": "bloom", "policy": "leftmost"})
+        elif mode == "both":
+            params.update({"deepreduce": "both", "index": "bloom", "value": "polyfit"})
+        elif mode == "dense":
+            params = {"compressor": "none", "memory": "none", "communicator": "allreduce"}
+        grc_a = deepreduce_from_params(dict(params))
+        grc_b = deepreduce_from_params(dict(params))
+
+        torch.manual_seed(300 + rank)
+        named = [("a", torch.randn(9000)), ("b", torch.randn(4, 700)),
+                 ("c", torch.randn(64))]
+        ok = True
+        for step in range(3):  # multiple steps: residuals must evolve identically
+            grads = [(n, (t * (step + 1)).clone()) for n, t in named]
+            fused = grc_a.step_many([(n, t.clone()) for n, t in grads])
+            loop = [grc_b.step(t.clone(), n) for n, t in grads]
+            for f, l in zip(fused, loop):
+                if not torch.allclose(f.reshape(-1), l.reshape(-1), atol=1e-5):
+                    ok = False
+        q.put((rank, ok))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+@pytest.mark.parametrize("mode", ["plain", "index", "both", "dense"])
+def test_fused_step_many_matches_per_tensor(mode):
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    os.environ["MASTER_PORT"] = str(29680 + hash("f" + mode) % 50)
+    procs = [ctx.Process(target=_run_fused_equivalence, args=(r, world, mode, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, ok in results:
+        assert ok is True, f"rank {rank}: {ok}"

@@ -15,7 +15,8 @@ Default fpr = 0.1 * num_indices / grad_size (pytorch/deepreduce.py:511).
 Policies (pytorch/deepreduce.py:479-492 + tensorflow/policies.hpp):
     'leftmost'      first K positives (ascending)
     'random'        K positives drawn with a fixed-seed permutation
-    'p0'            ALL positives; true count travels prepended to vals
+    'p0'            ALL positives; true count travels in-band at the tail
+                    of the bit array (4 bytes LE)
     'conflict_sets' one element per hash-conflict set, round-robin (parity
                     with policies.hpp:136-146; deterministic tie-breaks)
 
@@ -137,8 +138,14 @@ class Bloom(SparseCompressor):
             params["_own_decoded"] = (vals, new_idxs)
 
         if policy == "p0":
-            count = torch.as_tensor([num_indices], dtype=vals.dtype, device=vals.device)
-            vals = torch.cat([count, vals], dim=0)
+            # True insert count travels IN-BAND at the tail of the bit
+            # array (4 bytes LE) — the filter config (m, num_hash) is
+            # derived from it on decompress.  Keeping `vals` pure lets the
+            # 'both' wrapper feed them straight into the value codec
+            # (the reference prepended the count to vals,
+            # pytorch/deepreduce.py:525-527, which breaks exactly that).
+            count = torch.tensor([num_indices], dtype=torch.int32, device=packed.device)
+            packed = torch.cat([packed, count.view(torch.uint8)])
 
         return vals, packed, shape
 
@@ -147,8 +154,8 @@ class Bloom(SparseCompressor):
         vals, packed, shape = bf_sparse_tensor
         policy = params.get("policy", "leftmost")
         if policy == "p0":
-            count, vals = vals.split([1, vals.numel() - 1])
-            num_indices = int(count.item())
+            num_indices = int(packed[-4:].clone().view(torch.int32).item())
+            packed = packed[:-4]
         else:
             num_indices = int(vals.numel())
         grad_size = int(torch.Size(shape).numel())

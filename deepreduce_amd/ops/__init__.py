@@ -66,7 +66,8 @@ def topk_select(flat: torch.Tensor, k: int):
         flat.is_cuda
         and flat.dtype == torch.float32
         and 1 <= k <= flat.numel()
-        and flat.numel() >= 4096  # tiny tensors: torch.topk is fewer launches
+        and flat.numel() >= int(os.environ.get("DEEPREDUCE_RADIX_MIN", "4096"))
+        # tiny tensors: torch.topk is fewer launches
         and os.environ.get("DEEPREDUCE_TORCH_TOPK") != "1"
         and _want_hip(flat)
     ):

@@ -65,6 +65,18 @@ __host__ __device__ __forceinline__ bool bloom_test(const uint8_t* __restrict__ 
 // bloom insert: race-tolerant atomicOr bit sets (idempotent — paper App. E)
 // ---------------------------------------------------------------------------
 
+// Graph-safe zero fill (hipMemsetAsync becomes a memset NODE under stream
+// capture; a plain kernel keeps the captured graph homogeneous)
+__global__ void fill_zero_kernel(int* __restrict__ p, int64_t n) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) p[i] = 0;
+}
+static inline void zero_ints(int* p, int64_t n, hipStream_t s) {
+    int blocks = (int)min((long long)ceil_div(n, 256), 1024ll);
+    hipLaunchKernelGGL(fill_zero_kernel, dim3(blocks), dim3(256), 0, s, p, n);
+}
+
 __global__ void bloom_insert_kernel(const int64_t* __restrict__ items, int64_t n, int k,
                                     int64_t m, uint32_t* __restrict__ words) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -86,7 +98,7 @@ torch::Tensor bloom_insert(torch::Tensor idxs, int64_t m, int64_t num_hash) {
     auto words = torch::empty({nwords}, torch::dtype(torch::kInt32).device(idxs.device()));
     int64_t n = items.numel();
     hipStream_t stream = at::hip::getCurrentHIPStream();
-    (void)hipMemsetAsync(words.data_ptr<int32_t>(), 0, (size_t)nwords * 4, stream);
+    zero_ints(words.data_ptr<int32_t>(), nwords, stream);
     if (n > 0) {
         int threads = 256;
         int blocks = (int)std::min<int64_t>(ceil_div(n, threads), 4096);
@@ -328,7 +340,7 @@ torch::Tensor bloom_query_leftmost(torch::Tensor packed2d, int64_t m, int64_t nu
     hipLaunchKernelGGL(exclusive_scan_rows_kernel, dim3(R), dim3(QBLOCK), 0, stream,
                        counts, nblocks, offsets, (int*)nullptr);
     auto out = torch::empty({R * k_out}, torch::dtype(torch::kInt64).device(dev));
-    (void)hipMemsetAsync(out.data_ptr<int64_t>(), 0, (size_t)(R * k_out) * 8, stream);
+    zero_ints((int*)out.data_ptr<int64_t>(), 2 * R * k_out, stream);
     hipLaunchKernelGGL(bloom_scatter_kernel, dim3((int)nblocks), dim3(QBLOCK), 0, stream,
                        (const uint64_t*)mask.data_ptr<int64_t>(), mask_words, R, universe,
                        chunk, offsets, (const int64_t*)nullptr, k_out, k_out,
@@ -755,7 +767,7 @@ std::vector<torch::Tensor> topk_select(torch::Tensor flat, int64_t k) {
     int* sc = hist2 + TK_BINS;
     int* counts = sc + 4;
     int* offs = counts + 2 * nblocks;
-    (void)hipMemsetAsync(hist1, 0, (size_t)(2 * TK_BINS + 4) * sizeof(int), stream);
+    zero_ints(hist1, 2 * TK_BINS + 4, stream);
 
     int hblocks = (int)std::min<int64_t>(ceil_div(n, TK_BLOCK * 16), 2048);
     hipLaunchKernelGGL(topk_hist_kernel, dim3(hblocks), dim3(TK_BLOCK), 0, stream,

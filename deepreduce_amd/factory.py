@@ -42,6 +42,7 @@ def grace_from_params(params: dict):
     compressor = cls(**kwargs)
     memory = memory_registry[mem_name]()
     communicator = communicator_registry[comm_name](compressor, memory)
+    communicator.params = params  # runtime knobs (small_dense, ...)
     return communicator
 
 

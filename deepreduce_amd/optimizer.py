@@ -58,11 +58,15 @@ def _graph_safe(grc) -> bool:
     payload sizes: leftmost-policy bloom (sync-free query), qsgd, plain
     topk/randomk/none — NOT polyfit (segment count depends on num_pos),
     p0/conflict_sets/random policies, or threshold sparsification."""
-    # Opt-in while hipGraph replay is being qualified on real hardware;
-    # flipped to default-on once the soak run below is green (see
-    # profiles/NOTES.md).  DEEPREDUCE_GRAPH=0 always disables.
-    if os.environ.get("DEEPREDUCE_GRAPH", "0") != "1":
+    if os.environ.get("DEEPREDUCE_GRAPH", "1") == "0":
         return False
+    # Collectives inside hipGraph capture (RCCL) cannot be qualified on the
+    # single-GPU boxes we test on — multi-rank capture stays opt-in.
+    import torch.distributed as dist
+
+    if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+        if os.environ.get("DEEPREDUCE_GRAPH_DIST", "0") != "1":
+            return False
     comp = getattr(grc, "compressor", None)
     params = getattr(comp, "params", None)
     if params is None:  # bare sparsifier

@@ -177,6 +177,7 @@ def _run_fused_equivalence(rank, world, mode, q):
             "memory": "residual",
             "communicator": "allgather",
             "compress_ratio": 0.01,
+            "small_dense": False,  # strict per-tensor parity with grc.step
         }
         if mode == "index":
             params.update({"deepreduce": "index", "index": "bloom", "policy": "leftmost"})
@@ -212,6 +213,46 @@ def test_fused_step_many_matches_per_tensor(mode):
     os.environ["MASTER_PORT"] = str(29680 + hash("f" + mode) % 50)
     procs = [ctx.Process(target=_run_fused_equivalence, args=(r, world, mode, q))
              for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, ok in results:
+        assert ok is True, f"rank {rank}: {ok}"
+
+
+def _run_small_dense(rank, world, q):
+    try:
+        _init(rank, world)
+        from deepreduce_amd import deepreduce_from_params
+
+        grc = deepreduce_from_params({
+            "compressor": "topk", "memory": "residual",
+            "communicator": "allgather", "compress_ratio": 0.01,
+            "deepreduce": "index", "index": "bloom", "policy": "leftmost",
+        })
+        torch.manual_seed(500 + rank)
+        small = torch.randn(64)
+        large = torch.randn(8192)
+        out = grc.step_many([("w", large.clone()), ("b", small.clone())])
+        # small tensor must be the EXACT dense average across ranks
+        gathered = [torch.empty_like(small) for _ in range(world)]
+        dist.all_gather(gathered, small)
+        expect = torch.stack(gathered).mean(0)
+        ok = torch.allclose(out[1], expect, atol=1e-6)
+        q.put((rank, bool(ok)))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+def test_small_tensors_travel_dense_exact():
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    os.environ["MASTER_PORT"] = "29745"
+    procs = [ctx.Process(target=_run_small_dense, args=(r, world, q)) for r in range(world)]
     for p in procs:
         p.start()
     results = [q.get(timeout=180) for _ in range(world)]

@@ -72,25 +72,67 @@ class ResidualMemory(Memory):
             self.residuals[name] = tensor - decompressed.view_as(tensor)
 
     def compensate_many(self, tensors, names):
-        """Batched compensate: residual buffers are materialized (zeros) on
-        first sight so the math is a single _foreach op afterwards; buffers
-        inherit the gradient's layout (zeros_like) so channels-last convs
-        stay on the TensorIterator fast path."""
-        rs = []
-        for t, n in zip(tensors, names):
-            r = self.residuals.get(n)
-            if r is None or r.shape != t.shape:
-                r = torch.zeros_like(t)
-                self.residuals[n] = r
-            rs.append(r)
+        """Batched compensate over ONE flat residual pool.
+
+        All residuals live in a single flat float32 buffer (per-name
+        offsets); the whole-model compensate is one cat + one axpy over
+        ~25M elements (≈3 HBM passes, tens of µs at 8 TB/s) instead of
+        one kernel per tensor.  The flat buffer is persistent and updated
+        in place — hipGraph-stable.  Falls back to per-tensor math when
+        the tensor set is heterogeneous (mixed dtype/device).
+        """
+        key = tuple((n, tuple(t.shape)) for n, t in zip(names, tensors))
+        homogeneous = (
+            len({(t.dtype, t.device) for t in tensors}) == 1 and len(tensors) > 1
+        )
+        if not homogeneous:
+            rs = []
+            for t, n in zip(tensors, names):
+                r = self.residuals.get(n)
+                if r is None or r.shape != t.shape:
+                    r = torch.zeros_like(t)
+                    self.residuals[n] = r
+                rs.append(r)
+            if self.beta == 1.0 and self.gamma == 1.0:
+                return torch._foreach_add(rs, tensors)
+            out = torch._foreach_mul(tensors, self.gamma)
+            torch._foreach_add_(out, rs, alpha=self.beta)
+            return out
+
+        if getattr(self, "_flat_key", None) != key:
+            total = sum(t.numel() for t in tensors)
+            self._flat_r = torch.zeros(total, dtype=tensors[0].dtype,
+                                       device=tensors[0].device)
+            self._flat_key = key
+            self._flat_offsets = []
+            off = 0
+            for n, t in zip(names, tensors):
+                self._flat_offsets.append(off)
+                old = self.residuals.get(n)
+                view = self._flat_r[off : off + t.numel()].view(t.shape)
+                if old is not None and old.shape == t.shape:
+                    view.copy_(old.to(view.device, view.dtype))  # checkpoint resume
+                # expose per-name views so state_dict/checkpoint still works
+                self.residuals[n] = view
+                off += t.numel()
+        g_flat = torch.cat([t.reshape(-1) for t in tensors])
         if self.beta == 1.0 and self.gamma == 1.0:
-            return torch._foreach_add(rs, tensors)
-        out = torch._foreach_mul(tensors, self.gamma)
-        torch._foreach_add_(out, rs, alpha=self.beta)
+            c_flat = g_flat.add_(self._flat_r)  # g_flat is a fresh buffer
+        else:
+            c_flat = g_flat.mul_(self.gamma).add_(self._flat_r, alpha=self.beta)
+        self._flat_c = c_flat
+        out = []
+        for off, t in zip(self._flat_offsets, tensors):
+            out.append(c_flat[off : off + t.numel()].view(t.shape))
         return out
 
     def update_many(self, tensors, names, decompressed):
         """residual <- compensated - decompressed, in place (graph-stable)."""
+        if getattr(self, "_flat_key", None) is not None and len(tensors) > 1 \
+                and tensors[0].data_ptr() == self._flat_c.data_ptr():
+            d_flat = torch.cat([d.reshape(-1) for d in decompressed])
+            torch.sub(self._flat_c, d_flat, out=self._flat_r)
+            return
         rs = [self.residuals[n] for n in names]
         torch._foreach_copy_(rs, list(tensors))
         torch._foreach_sub_(rs, [d.view_as(t) for d, t in zip(decompressed, tensors)])
@@ -104,6 +146,7 @@ class ResidualMemory(Memory):
         self.residuals = state["residuals"]
         self.beta = state["beta"]
         self.gamma = state["gamma"]
+        self._flat_key = None  # rebuild the flat pool from loaded values
 
 
 memory_registry = {

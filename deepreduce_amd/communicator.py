@@ -177,6 +177,11 @@ class Allgather(Communicator):
 
     def _step_many_impl(self, named_tensors):
         comp = self.compressor
+        from .ops import batched as _bt
+
+        bp = _bt.maybe_pipeline(self, comp, named_tensors)
+        if bp is not None:
+            return self._step_many_batched(bp, named_tensors)
         names = [n for n, _ in named_tensors]
         grads = [t for _, t in named_tensors]
         compensated = self.memory.compensate_many(grads, names)
@@ -228,6 +233,60 @@ class Allgather(Communicator):
         if comp.average:
             torch._foreach_div_(totals, world)
         return [t.view_as(g) for t, g in zip(totals, grads)]
+
+    def _step_many_batched(self, bp, named_tensors):
+        """Whole-model fused pipeline: ~12 kernels + ONE collective per step
+        regardless of tensor count (ops/batched.py; bt_* kernels)."""
+        from .memory import ResidualMemory
+
+        names = [n for n, _ in named_tensors]
+        grads = [t for _, t in named_tensors]
+        mem = self.memory
+        compensated = mem.compensate_many(grads, names)
+        flat_c = getattr(mem, "_flat_c", None)
+        if (flat_c is not None and compensated
+                and compensated[0].data_ptr() == flat_c.data_ptr()
+                and flat_c.numel() == bp.total_values):
+            c_flat = flat_c  # flat residual pool already concatenated them
+        else:
+            c_flat = torch.cat([t.reshape(-1) for t in compensated])
+
+        wire, out_idx = bp.compress(c_flat)
+        self.last_wire_bytes = int(wire.numel())
+        own_dense = bp.decode_own(wire, out_idx)
+
+        # residual <- compensated - own decode (exactly the generic math)
+        if isinstance(mem, ResidualMemory):
+            if c_flat is flat_c:
+                torch.sub(c_flat, own_dense, out=mem._flat_r)
+            else:
+                offs = 0
+                decs = []
+                for t in grads:
+                    decs.append(own_dense[offs : offs + t.numel()].view(t.shape))
+                    offs += t.numel()
+                mem.update_many(compensated, names, decs)
+
+        world = self.world_size
+        if world == 1:
+            result = own_dense
+        else:
+            gathered = torch.empty(world, wire.numel(), dtype=torch.uint8,
+                                   device=wire.device)
+            try:
+                dist.all_gather_into_tensor(gathered.view(-1), wire)
+            except (AttributeError, RuntimeError):
+                bufs = list(gathered.unbind(0))
+                dist.all_gather(bufs, wire)
+            result = bp.decode_sum(gathered)
+            if self.compressor.average:
+                result /= world
+        outs = []
+        offs = 0
+        for t in grads:
+            outs.append(result[offs : offs + t.numel()].view(t.shape))
+            offs += t.numel()
+        return outs
 
     @staticmethod
     def _slice_buffer(big, sizes, metas):

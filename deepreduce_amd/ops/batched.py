@@ -1,0 +1,132 @@
+"""Whole-model batched compression pipeline driver (MI355X hot path).
+
+Builds the static per-model descriptor table consumed by the bt_* kernels
+in ops/src/hip_ops.hip: every kernel walks a block->tensor map so all
+tensors' chunks execute concurrently, and the compress side writes the
+fused wire buffer (byte-identical to communicator._flatten_payload of the
+per-tensor path: per tensor [float32 vals (k, padded to 8B)] [uint8 bloom
+bits (ceil(m/8), padded to 8B)]) in ~12 kernels for the entire model.
+
+Applies to the flagship configuration: top-k sparsifier + Bloom index codec
++ leftmost policy (sync-free), float32 CUDA gradients, tensors above the
+1000-element codec bypass.  Anything else falls back to the generic
+per-tensor path in communicator.step_many.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+
+BT_CHUNK = 8192
+
+
+def _pad8(x: int) -> int:
+    return (x + 7) & ~7
+
+
+class BatchedPipeline:
+    def __init__(self, names, numels, params, device):
+        from ..codecs.bloom import Bloom
+
+        self.names = list(names)
+        self.numels = list(numels)
+        T = len(numels)
+        ratio = params.get("compress_ratio", 0.01)
+
+        desc = torch.zeros(T, 12, dtype=torch.int64)
+        b2t = []
+        voff = koff = wire_off = cntoff = mwoff = blkoff = 0
+        self.metas = []
+        for t, n in enumerate(numels):
+            k = max(1, int(round(n * ratio)))
+            num_hash, m = Bloom._config(k, n, params)
+            nbytes = (m + 7) // 8
+            nb = (n + BT_CHUNK - 1) // BT_CHUNK
+            desc[t, 0] = n
+            desc[t, 1] = voff
+            desc[t, 2] = k
+            desc[t, 3] = koff
+            desc[t, 4] = m
+            desc[t, 5] = num_hash
+            desc[t, 6] = wire_off + _pad8(4 * k)   # bits after padded vals
+            desc[t, 7] = wire_off                  # vals first (payload order)
+            desc[t, 8] = cntoff
+            desc[t, 9] = mwoff
+            desc[t, 10] = blkoff
+            self.metas.append([(torch.float32, k), (torch.uint8, nbytes)])
+            b2t.extend([t] * nb)
+            voff += n
+            koff += k
+            wire_off += _pad8(4 * k) + _pad8(nbytes)
+            cntoff += nb
+            mwoff += nb * (BT_CHUNK // 64)
+            blkoff += nb
+
+        self.total_values = voff
+        self.k_total = koff
+        self.wire_bytes = wire_off          # multiple of 8 by construction
+        self.mask_words = mwoff
+        self.desc = desc.to(device)
+        self.b2t = torch.tensor(b2t, dtype=torch.int32, device=device)
+
+    # -- kernels ----------------------------------------------------------
+    def compress(self, values_flat: torch.Tensor):
+        from deepreduce_amd import _hip_ops
+
+        wire, out_idx = _hip_ops.batched_compress(
+            values_flat, self.desc, self.b2t, self.wire_bytes, self.k_total,
+            self.mask_words,
+        )
+        return wire, out_idx
+
+    def decode_own(self, wire, out_idx):
+        from deepreduce_amd import _hip_ops
+
+        return _hip_ops.batched_scatter_dense(wire, out_idx, self.desc,
+                                              self.total_values)
+
+    def decode_sum(self, wires2d):
+        from deepreduce_amd import _hip_ops
+
+        return _hip_ops.batched_decode_sum(wires2d, self.desc, self.b2t,
+                                           self.total_values, self.mask_words)
+
+
+def maybe_pipeline(communicator, comp, named_tensors):
+    """Return a cached BatchedPipeline when the configuration qualifies."""
+    from . import hip_available
+    from ..codecs.bloom import Bloom
+    from ..compressors import TopKCompressor
+    from ..wrappers import IndexCompressor
+
+    if not isinstance(comp, IndexCompressor) or comp.idx_codec is not Bloom:
+        return None
+    params = comp.params
+    if params.get("policy", "leftmost") != "leftmost" or params.get("micro-benchmark"):
+        return None
+    if not isinstance(comp.sparsifier, TopKCompressor):
+        return None
+    grads = [t for _, t in named_tensors]
+    if not grads or not grads[0].is_cuda or not hip_available():
+        return None
+    if any(t.dtype != torch.float32 or not t.is_cuda for t in grads):
+        return None
+    if any(t.numel() <= 1000 for t in grads):  # codec bypass: generic path
+        return None
+    if communicator.world_size > 16:
+        return None
+    key = (
+        tuple(n for n, _ in named_tensors),
+        tuple(t.numel() for t in grads),
+        comp.sparsifier.compress_ratio,
+        params.get("fpr"),
+        str(grads[0].device),
+    )
+    cached = getattr(communicator, "_bt_pipeline", None)
+    if cached is not None and cached[0] == key:
+        return cached[1]
+    bp = BatchedPipeline([n for n, _ in named_tensors],
+                         [t.numel() for t in grads], params, grads[0].device)
+    communicator._bt_pipeline = (key, bp)
+    return bp

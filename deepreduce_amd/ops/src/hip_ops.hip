@@ -855,6 +855,608 @@ torch::Tensor cholesky_solve_small(torch::Tensor G, torch::Tensor b) {
     return x;
 }
 
+
+// ===========================================================================
+// Batched cross-tensor pipeline (the MI355X-first hot path).
+//
+// One training step compresses ~54 large tensors.  Launched per tensor, the
+// radix-select/bloom kernels are each a few microseconds of real work — the
+// step becomes kernel-count bound even under hipGraph replay (profiles/
+// NOTES.md r03).  Here the WHOLE model is processed by ~17 kernels total:
+// every kernel walks a block->tensor map (b2t) so all tensors' chunks fill
+// the 256 CUs concurrently, and the compress side writes the fused wire
+// buffer (the exact _flatten_payload layout) directly — no torch-op glue,
+// no cats, no per-tensor launches.
+//
+// Descriptor row (int64 [T, 12], built once per model layout in
+// deepreduce_amd/ops/batched.py and cached on device):
+//   0 n        elements in tensor
+//   1 voff     offset into the flat values buffer
+//   2 k        top-k count
+//   3 koff     prefix sum of k (output offset)
+//   4 m        bloom bits
+//   5 nh       bloom hash count
+//   6 bitoff   absolute byte offset of the bloom bit array in the wire
+//   7 valoff   absolute byte offset of the FP-aware values in the wire
+//   8 cntoff   prefix sum of per-tensor block counts (count/offset slots)
+//   9 mwoff    prefix sum of per-tensor ballot-mask words
+//  10 blkoff   prefix sum of block indices (global block -> local block)
+//  11 reserved
+// ===========================================================================
+
+#define BT_CHUNK 8192  // elements per block (multiple of 256)
+#define BT_F 12
+
+__device__ __forceinline__ const int64_t* bt_row(const int64_t* desc, int t) {
+    return desc + (int64_t)t * BT_F;
+}
+
+// level 1: 11-bit high-key histogram per tensor; level 2: low 11 bits of the
+// level-1 tie bin.
+__global__ void bt_hist_kernel(const float* __restrict__ vals,
+                               const int64_t* __restrict__ desc,
+                               const int* __restrict__ b2t, int level,
+                               int* __restrict__ hist /*[T,2048] plane*/,
+                               const int* __restrict__ sc /*[T,4]*/) {
+    const int t = b2t[blockIdx.x];
+    const int64_t* D = bt_row(desc, t);
+    const int64_t lb = blockIdx.x - D[10];
+    const int64_t start = lb * BT_CHUNK;
+    const int64_t end = min(start + BT_CHUNK, D[0]);
+    const float* __restrict__ v = vals + D[1];
+    __shared__ int lh[TK_BINS];
+    for (int i = threadIdx.x; i < TK_BINS; i += blockDim.x) lh[i] = 0;
+    __syncthreads();
+    if (level == 1) {
+        for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+            atomicAdd(&lh[tk_key22(v[i]) >> 11], 1);
+    } else {
+        const uint32_t bstar = (uint32_t)sc[t * 4 + 0];
+        for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+            uint32_t key = tk_key22(v[i]);
+            if ((key >> 11) == bstar) atomicAdd(&lh[key & 2047u], 1);
+        }
+    }
+    __syncthreads();
+    int* H = hist + (int64_t)t * TK_BINS;
+    for (int i = threadIdx.x; i < TK_BINS; i += blockDim.x)
+        if (lh[i]) atomicAdd(&H[i], lh[i]);
+}
+
+// one wavefront per tensor: derive the threshold bin from the histogram
+// (same math as topk_thresh_kernel, batched over T)
+__global__ void bt_thresh_kernel(const int* __restrict__ hist,
+                                 const int64_t* __restrict__ desc, int level,
+                                 int* __restrict__ sc) {
+    const int t = blockIdx.x;
+    const int lane = threadIdx.x;
+    const int64_t k = bt_row(desc, t)[2];
+    const int* H = hist + (int64_t)t * TK_BINS;
+    int* S = sc + t * 4;
+    const int target = (level == 1) ? (int)k : (int)k - S[1];
+    const int SEG = TK_BINS / WAVE;
+    int h[SEG];
+    int seg_sum = 0;
+    for (int j = 0; j < SEG; ++j) {
+        h[j] = H[lane * SEG + j];
+        seg_sum += h[j];
+    }
+    int suffix_excl;
+    {
+        int acc = seg_sum;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int other = __shfl_down(acc, off, WAVE);
+            if (lane + off < WAVE) acc += other;
+        }
+        suffix_excl = acc - seg_sum;
+    }
+    int found_b = -1, found_above = 0;
+    int acc = 0;
+    for (int j = SEG - 1; j >= 0; --j) {
+        int suffix_here = suffix_excl + acc + h[j];
+        if (suffix_here >= target) {
+            found_b = lane * SEG + j;
+            found_above = suffix_excl + acc;
+            break;
+        }
+        acc += h[j];
+    }
+    int best = found_b;
+    for (int off = 1; off < WAVE; off <<= 1) {
+        int other = __shfl_down(best, off, WAVE);
+        if (lane + off < WAVE && other > best) best = other;
+    }
+    best = __shfl(best, 0, WAVE);
+    if (found_b == best && found_b >= 0) {
+        if (level == 1) {
+            S[0] = found_b;
+            S[1] = found_above;
+        } else {
+            S[2] = S[0] * TK_BINS + found_b;
+            S[3] = S[1] + found_above;
+        }
+    }
+}
+
+// per-block {key > T} / {key == T} counts -> counts planes [2, BV]
+__global__ void bt_count_kernel(const float* __restrict__ vals,
+                                const int64_t* __restrict__ desc,
+                                const int* __restrict__ b2t,
+                                const int* __restrict__ sc, int64_t BV,
+                                int* __restrict__ counts) {
+    const int t = b2t[blockIdx.x];
+    const int64_t* D = bt_row(desc, t);
+    const int64_t lb = blockIdx.x - D[10];
+    const int64_t start = lb * BT_CHUNK;
+    const int64_t end = min(start + BT_CHUNK, D[0]);
+    const float* __restrict__ v = vals + D[1];
+    const uint32_t T22 = (uint32_t)sc[t * 4 + 2];
+    int c0 = 0, c1 = 0;
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+        uint32_t key = tk_key22(v[i]);
+        c0 += (key > T22);
+        c1 += (key == T22);
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        c0 += __shfl_down(c0, off, WAVE);
+        c1 += __shfl_down(c1, off, WAVE);
+    }
+    __shared__ int w0[TK_BLOCK / WAVE], w1[TK_BLOCK / WAVE];
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) { w0[wid] = c0; w1[wid] = c1; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int t0 = 0, t1 = 0;
+        for (int w = 0; w < TK_BLOCK / WAVE; ++w) { t0 += w0[w]; t1 += w1[w]; }
+        counts[D[8] + lb] = t0;
+        counts[BV + D[8] + lb] = t1;
+    }
+}
+
+// generic device-side exclusive row scan used by the batched scans
+__device__ void bt_scan_row(const int* __restrict__ src, int64_t n,
+                            int* __restrict__ dst) {
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    __shared__ int wave_tot[QBLOCK / WAVE];
+    __shared__ int carry;
+    if (threadIdx.x == 0) carry = 0;
+    __syncthreads();
+    for (int64_t i0 = 0; i0 < n; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        int v = (i < n) ? src[i] : 0;
+        int incl = v;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        if (lane == WAVE - 1) wave_tot[wid] = incl;
+        __syncthreads();
+        int wbase = 0;
+        for (int w = 0; w < wid; ++w) wbase += wave_tot[w];
+        if (i < n) dst[i] = carry + wbase + incl - v;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int tsum = 0;
+            for (int w = 0; w < QBLOCK / WAVE; ++w) tsum += wave_tot[w];
+            carry += tsum;
+        }
+        __syncthreads();
+    }
+}
+
+// grid = planes*T blocks: scan tensor t's slot row within each plane
+__global__ void bt_scan_kernel(const int* __restrict__ counts,
+                               const int64_t* __restrict__ desc, int nT,
+                               int64_t BV, int* __restrict__ offs) {
+    const int t = blockIdx.x % nT;
+    const int64_t plane = blockIdx.x / nT;
+    const int64_t* D = bt_row(desc, t);
+    const int64_t nb = (D[0] + BT_CHUNK - 1) / BT_CHUNK;
+    bt_scan_row(counts + plane * BV + D[8], nb, offs + plane * BV + D[8]);
+}
+
+// ordered two-class compaction -> out_i only (values are re-read FP-aware
+// from the dense tensor at the bloom-recovered positions later)
+__global__ void bt_scatter_kernel(const float* __restrict__ vals,
+                                  const int64_t* __restrict__ desc,
+                                  const int* __restrict__ b2t,
+                                  const int* __restrict__ sc,
+                                  const int* __restrict__ offs, int64_t BV,
+                                  int64_t* __restrict__ out_i) {
+    const int t = b2t[blockIdx.x];
+    const int64_t* D = bt_row(desc, t);
+    const int64_t lb = blockIdx.x - D[10];
+    const int64_t start = lb * BT_CHUNK;
+    const int64_t end = min(start + BT_CHUNK, D[0]);
+    const float* __restrict__ v = vals + D[1];
+    const uint32_t T22 = (uint32_t)sc[t * 4 + 2];
+    const int64_t above = sc[t * 4 + 3];
+    const int64_t k = D[2];
+    const int64_t need = k - above;
+    int64_t* __restrict__ out = out_i + D[3];
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    __shared__ int wc0[TK_BLOCK / WAVE], wc1[TK_BLOCK / WAVE];
+    __shared__ int base0_s, base1_s;
+    if (threadIdx.x == 0) {
+        base0_s = offs[D[8] + lb];
+        base1_s = offs[BV + D[8] + lb];
+    }
+    __syncthreads();
+    for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        bool in_r = (i < end);
+        uint32_t key = in_r ? tk_key22(v[i]) : 0u;
+        bool p0 = in_r && (key > T22);
+        bool p1 = in_r && (key == T22);
+        uint64_t b0 = __ballot(p0), b1 = __ballot(p1);
+        if (lane == 0) { wc0[wid] = __popcll(b0); wc1[wid] = __popcll(b1); }
+        __syncthreads();
+        uint64_t below = (lane == 63) ? (~0ull >> 1) : ((1ull << lane) - 1);
+        if (p0) {
+            int wbase = 0;
+            for (int w = 0; w < wid; ++w) wbase += wc0[w];
+            out[base0_s + wbase + __popcll(b0 & below)] = i;
+        } else if (p1) {
+            int wbase = 0;
+            for (int w = 0; w < wid; ++w) wbase += wc1[w];
+            int64_t ordinal = base1_s + wbase + __popcll(b1 & below);
+            if (ordinal < need) out[above + ordinal] = i;
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int t0 = 0, t1 = 0;
+            for (int w = 0; w < TK_BLOCK / WAVE; ++w) { t0 += wc0[w]; t1 += wc1[w]; }
+            base0_s += t0;
+            base1_s += t1;
+        }
+        __syncthreads();
+    }
+}
+
+// hash every selected index into its tensor's bloom bit array in the wire
+__global__ void bt_insert_kernel(const int64_t* __restrict__ out_i,
+                                 const int64_t* __restrict__ desc, int nT,
+                                 int64_t K, uint8_t* __restrict__ wire) {
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < K; j += stride) {
+        // binary search: largest t with koff[t] <= j
+        int lo = 0, hi = nT - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
+        }
+        const int64_t* D = bt_row(desc, lo);
+        uint32_t h1, h2;
+        hash_bases(out_i[j], &h1, &h2);
+        const int64_t m = D[4];
+        uint32_t* bits = (uint32_t*)(wire + D[6]);
+        uint64_t pos = (uint64_t)h1 % (uint64_t)m;
+        uint64_t step = (uint64_t)h2 % (uint64_t)m;
+        for (int h = 0; h < (int)D[5]; ++h) {
+            atomicOr(&bits[pos >> 5], 1u << (pos & 31));
+            pos += step;
+            if (pos >= (uint64_t)m) pos -= (uint64_t)m;
+        }
+    }
+}
+
+// universe query over R wire buffers (filters share probe positions):
+// counts plane per rank + ballot bit-plane per rank
+__global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
+                                 int64_t wstride, int R,
+                                 const int64_t* __restrict__ desc,
+                                 const int* __restrict__ b2t, int64_t BV,
+                                 int64_t MW, int* __restrict__ qcounts,
+                                 uint64_t* __restrict__ mask) {
+    const int t = b2t[blockIdx.x];
+    const int64_t* D = bt_row(desc, t);
+    const int64_t lb = blockIdx.x - D[10];
+    const int64_t start = lb * BT_CHUNK;
+    const int64_t end = min(start + BT_CHUNK, D[0]);
+    const int64_t m = D[4];
+    const int nh = (int)D[5];
+    const int64_t bitoff = D[6];
+    const int64_t mwoff = D[9];
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    int cnt[MAXR];
+    for (int r = 0; r < R; ++r) cnt[r] = 0;
+    for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        unsigned alive = (i < end) ? ((1u << R) - 1) : 0u;
+        if (alive) {
+            uint32_t h1, h2;
+            hash_bases(i, &h1, &h2);
+            uint64_t pos = (uint64_t)h1 % (uint64_t)m;
+            uint64_t step = (uint64_t)h2 % (uint64_t)m;
+            for (int h = 0; h < nh && alive; ++h) {
+                int64_t byte = pos >> 3;
+                uint8_t bit = pos & 7;
+                for (int r = 0; r < R; ++r)
+                    if (alive & (1u << r))
+                        if (!((wires[r * wstride + bitoff + byte] >> bit) & 1))
+                            alive &= ~(1u << r);
+                pos += step;
+                if (pos >= (uint64_t)m) pos -= (uint64_t)m;
+            }
+        }
+        for (int r = 0; r < R; ++r) {
+            uint64_t ball = __ballot(alive & (1u << r));
+            cnt[r] += __popcll(ball);
+            if (lane == 0)
+                mask[r * MW + mwoff + ((i0 + (int64_t)wid * WAVE) >> 6)] = ball;
+        }
+    }
+    __shared__ int wsum[MAXR][QBLOCK / WAVE];
+    if (lane == 0)
+        for (int r = 0; r < R; ++r) wsum[r][wid] = cnt[r];
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int r = 0; r < R; ++r) {
+            int total = 0;
+            for (int w = 0; w < QBLOCK / WAVE; ++w) total += wsum[r][w];
+            qcounts[r * BV + D[8] + lb] = total;
+        }
+    }
+}
+
+// compress-side compaction (R=1): leftmost-k indices -> out_idx, AND the
+// FP-aware value gather fused in: wire values = dense values at the
+// positions decompress will re-derive
+__global__ void bt_qscatter_own_kernel(const uint64_t* __restrict__ mask,
+                                       const int* __restrict__ qoffs,
+                                       const float* __restrict__ vals,
+                                       const int64_t* __restrict__ desc,
+                                       const int* __restrict__ b2t,
+                                       uint8_t* __restrict__ wire,
+                                       int64_t* __restrict__ out_idx) {
+    const int t = b2t[blockIdx.x];
+    const int64_t* D = bt_row(desc, t);
+    const int64_t lb = blockIdx.x - D[10];
+    const int64_t start = lb * BT_CHUNK;
+    const int64_t end = min(start + BT_CHUNK, D[0]);
+    const int64_t k = D[2];
+    const int64_t mwoff = D[9];
+    const float* __restrict__ v = vals + D[1];
+    float* __restrict__ wv = (float*)(wire + D[7]);
+    int64_t* __restrict__ oi = out_idx + D[3];
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    __shared__ int wave_cnt[QBLOCK / WAVE];
+    __shared__ int base_s;
+    if (threadIdx.x == 0) base_s = qoffs[D[8] + lb];
+    __syncthreads();
+    for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        uint64_t ball = mask[mwoff + ((i0 + (int64_t)wid * WAVE) >> 6)];
+        bool pred = (ball >> lane) & 1;
+        if (lane == 0) wave_cnt[wid] = __popcll(ball);
+        __syncthreads();
+        if (pred) {
+            int wbase = 0;
+            for (int w = 0; w < wid; ++w) wbase += wave_cnt[w];
+            uint64_t below = (lane == 63) ? (~0ull >> 1) : ((1ull << lane) - 1);
+            int64_t ord = base_s + wbase + __popcll(ball & below);
+            if (ord < k) {
+                oi[ord] = i;
+                wv[ord] = v[i];
+            }
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int tsum = 0;
+            for (int w = 0; w < QBLOCK / WAVE; ++w) tsum += wave_cnt[w];
+            base_s += tsum;
+        }
+        __syncthreads();
+    }
+}
+
+// decode-side compaction for rank r (launched per rank, sequentially, so the
+// dense accumulation order is deterministic and identical on every rank):
+// dense[voff + i] += wire_r values at the leftmost-k ordinals
+__global__ void bt_qscatter_add_kernel(const uint64_t* __restrict__ mask,
+                                       const int* __restrict__ qoffs, int r,
+                                       int64_t BV, int64_t MW,
+                                       const uint8_t* __restrict__ wires,
+                                       int64_t wstride,
+                                       const int64_t* __restrict__ desc,
+                                       const int* __restrict__ b2t,
+                                       float* __restrict__ dense) {
+    const int t = b2t[blockIdx.x];
+    const int64_t* D = bt_row(desc, t);
+    const int64_t lb = blockIdx.x - D[10];
+    const int64_t start = lb * BT_CHUNK;
+    const int64_t end = min(start + BT_CHUNK, D[0]);
+    const int64_t k = D[2];
+    const int64_t mwoff = D[9];
+    const float* __restrict__ wv = (const float*)(wires + r * wstride + D[7]);
+    float* __restrict__ dv = dense + D[1];
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    __shared__ int wave_cnt[QBLOCK / WAVE];
+    __shared__ int base_s;
+    if (threadIdx.x == 0) base_s = qoffs[r * BV + D[8] + lb];
+    __syncthreads();
+    for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        uint64_t ball = mask[r * MW + mwoff + ((i0 + (int64_t)wid * WAVE) >> 6)];
+        bool pred = (ball >> lane) & 1;
+        if (lane == 0) wave_cnt[wid] = __popcll(ball);
+        __syncthreads();
+        if (pred) {
+            int wbase = 0;
+            for (int w = 0; w < wid; ++w) wbase += wave_cnt[w];
+            uint64_t below = (lane == 63) ? (~0ull >> 1) : ((1ull << lane) - 1);
+            int64_t ord = base_s + wbase + __popcll(ball & below);
+            if (ord < k) dv[i] += wv[ord];
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int tsum = 0;
+            for (int w = 0; w < QBLOCK / WAVE; ++w) tsum += wave_cnt[w];
+            base_s += tsum;
+        }
+        __syncthreads();
+    }
+}
+
+// own-payload decode: dense[voff + out_idx] = wire values (indices unique)
+__global__ void bt_scatter_dense_kernel(const int64_t* __restrict__ out_idx,
+                                        const uint8_t* __restrict__ wire,
+                                        const int64_t* __restrict__ desc,
+                                        int nT, int64_t K,
+                                        float* __restrict__ dense) {
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < K; j += stride) {
+        int lo = 0, hi = nT - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
+        }
+        const int64_t* D = bt_row(desc, lo);
+        const float* wv = (const float*)(wire + D[7]);
+        dense[D[1] + out_idx[j]] = wv[j - D[3]];
+    }
+}
+
+__global__ void bt_fill_zero_f(float* __restrict__ p, int64_t n) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) p[i] = 0.f;
+}
+
+// ---------------------------------------------------------------------------
+// batched drivers
+// ---------------------------------------------------------------------------
+
+static inline int bt_grid(int64_t n) {
+    return (int)std::min<int64_t>(ceil_div(n, 256), 4096);
+}
+
+// Compress all tensors: flat values -> (wire, out_idx).  Every size below is
+// host-static for a fixed model layout, so the whole call is sync-free and
+// hipGraph-capturable.
+std::vector<torch::Tensor> batched_compress(torch::Tensor values_flat,
+                                            torch::Tensor desc,
+                                            torch::Tensor b2t,
+                                            int64_t wire_bytes, int64_t k_total,
+                                            int64_t mask_words) {
+    CHECK_CUDA(values_flat);
+    auto v = values_flat.contiguous();
+    auto d = desc.contiguous();
+    auto map = b2t.contiguous();
+    const int T = (int)d.size(0);
+    const int64_t BV = map.numel();
+    auto dev = v.device();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+
+    auto ws = torch::empty({(int64_t)T * (2 * TK_BINS + 4) + 6 * BV},
+                           torch::dtype(torch::kInt32).device(dev));
+    int* hist1 = ws.data_ptr<int>();
+    int* hist2 = hist1 + (int64_t)T * TK_BINS;
+    int* sc = hist2 + (int64_t)T * TK_BINS;
+    int* counts = sc + (int64_t)T * 4;   // 2 planes
+    int* offs = counts + 2 * BV;         // 2 planes
+    int* qcounts = offs + 2 * BV;
+    int* qoffs = qcounts + BV;
+    auto mask = torch::empty({mask_words}, torch::dtype(torch::kInt64).device(dev));
+    auto wire = torch::empty({wire_bytes}, torch::dtype(torch::kUInt8).device(dev));
+    auto out_idx = torch::empty({k_total}, torch::dtype(torch::kInt64).device(dev));
+
+    zero_ints(hist1, (int64_t)T * (2 * TK_BINS + 4), stream);
+    zero_ints((int*)wire.data_ptr<uint8_t>(), wire_bytes / 4, stream);  // wire_bytes % 8 == 0
+
+    const int64_t* dp = d.data_ptr<int64_t>();
+    const int* mp = map.data_ptr<int>();
+    const float* vp = v.data_ptr<float>();
+
+    hipLaunchKernelGGL(bt_hist_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, 1, hist1, sc);
+    hipLaunchKernelGGL(bt_thresh_kernel, dim3(T), dim3(WAVE), 0, stream, hist1, dp, 1, sc);
+    hipLaunchKernelGGL(bt_hist_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, 2, hist2, sc);
+    hipLaunchKernelGGL(bt_thresh_kernel, dim3(T), dim3(WAVE), 0, stream, hist2, dp, 2, sc);
+    hipLaunchKernelGGL(bt_count_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, sc, BV, counts);
+    hipLaunchKernelGGL(bt_scan_kernel, dim3(2 * T), dim3(QBLOCK), 0, stream,
+                       counts, dp, T, BV, offs);
+    hipLaunchKernelGGL(bt_scatter_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, sc, offs, BV, out_idx.data_ptr<int64_t>());
+    hipLaunchKernelGGL(bt_insert_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                       out_idx.data_ptr<int64_t>(), dp, T, k_total,
+                       wire.data_ptr<uint8_t>());
+    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+                       wire.data_ptr<uint8_t>(), wire_bytes, 1, dp, mp, BV,
+                       mask_words, qcounts, (uint64_t*)mask.data_ptr<int64_t>());
+    hipLaunchKernelGGL(bt_scan_kernel, dim3(T), dim3(QBLOCK), 0, stream,
+                       qcounts, dp, T, BV, qoffs);
+    hipLaunchKernelGGL(bt_qscatter_own_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+                       (const uint64_t*)mask.data_ptr<int64_t>(), qoffs, vp, dp, mp,
+                       wire.data_ptr<uint8_t>(), out_idx.data_ptr<int64_t>());
+    return {wire, out_idx};
+}
+
+// Own-payload decode: (wire, out_idx) -> dense flat
+torch::Tensor batched_scatter_dense(torch::Tensor wire, torch::Tensor out_idx,
+                                    torch::Tensor desc, int64_t total_values) {
+    CHECK_CUDA(wire);
+    auto d = desc.contiguous();
+    const int T = (int)d.size(0);
+    const int64_t K = out_idx.numel();
+    auto dense = torch::empty({total_values},
+                              torch::dtype(torch::kFloat32).device(wire.device()));
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(bt_fill_zero_f, dim3(bt_grid(total_values)), dim3(256), 0, stream,
+                       dense.data_ptr<float>(), total_values);
+    hipLaunchKernelGGL(bt_scatter_dense_kernel, dim3(bt_grid(K)), dim3(256), 0, stream,
+                       out_idx.data_ptr<int64_t>(), wire.data_ptr<uint8_t>(),
+                       d.data_ptr<int64_t>(), T, K, dense.data_ptr<float>());
+    return dense;
+}
+
+// Multi-rank decode: stacked wires [R, W] -> SUM of dense decodes.
+// Hashing runs once for all R filters; per-rank scatter-adds are launched
+// sequentially so the accumulation order is deterministic on every rank.
+torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
+                                 torch::Tensor b2t, int64_t total_values,
+                                 int64_t mask_words) {
+    CHECK_CUDA(wires2d);
+    TORCH_CHECK(wires2d.dim() == 2, "expected [R, W]");
+    auto w = wires2d.contiguous();
+    auto d = desc.contiguous();
+    auto map = b2t.contiguous();
+    const int R = (int)w.size(0);
+    TORCH_CHECK(R >= 1 && R <= MAXR, "1..16 ranks supported");
+    const int64_t W = w.size(1);
+    const int T = (int)d.size(0);
+    const int64_t BV = map.numel();
+    auto dev = w.device();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+
+    auto ws = torch::empty({2 * (int64_t)R * BV}, torch::dtype(torch::kInt32).device(dev));
+    int* qcounts = ws.data_ptr<int>();
+    int* qoffs = qcounts + (int64_t)R * BV;
+    auto mask = torch::empty({(int64_t)R * mask_words},
+                             torch::dtype(torch::kInt64).device(dev));
+    auto dense = torch::empty({total_values}, torch::dtype(torch::kFloat32).device(dev));
+    hipLaunchKernelGGL(bt_fill_zero_f, dim3(bt_grid(total_values)), dim3(256), 0, stream,
+                       dense.data_ptr<float>(), total_values);
+    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+                       w.data_ptr<uint8_t>(), W, R, d.data_ptr<int64_t>(),
+                       map.data_ptr<int>(), BV, mask_words, qcounts,
+                       (uint64_t*)mask.data_ptr<int64_t>());
+    hipLaunchKernelGGL(bt_scan_kernel, dim3(R * T), dim3(QBLOCK), 0, stream,
+                       qcounts, d.data_ptr<int64_t>(), T, BV, qoffs);
+    for (int r = 0; r < R; ++r) {
+        hipLaunchKernelGGL(bt_qscatter_add_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+                           (const uint64_t*)mask.data_ptr<int64_t>(), qoffs, r, BV,
+                           mask_words, w.data_ptr<uint8_t>(), W,
+                           d.data_ptr<int64_t>(), map.data_ptr<int>(),
+                           dense.data_ptr<float>());
+    }
+    return dense;
+}
+
 // ---------------------------------------------------------------------------
 // CPU-native C++ paths (replace the reference's TF C++ CPU ops:
 // bloom_filter_compression.cc / integer_compression.cc) — same wire format
@@ -978,6 +1580,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("unpack_ints", &unpack_ints, "n-bit unpack (HIP)");
     m.def("topk_select", &topk_select, "histogram-threshold top-k (HIP)");
     m.def("cholesky_solve_small", &cholesky_solve_small, "batched tiny SPD solve (HIP)");
+    m.def("batched_compress", &batched_compress,
+          "whole-model fused topk+bloom+query+gather -> (wire, out_idx)");
+    m.def("batched_scatter_dense", &batched_scatter_dense,
+          "own-payload decode: (wire, out_idx) -> dense flat");
+    m.def("batched_decode_sum", &batched_decode_sum,
+          "multi-rank decode: [R, W] wires -> sum of dense decodes");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
     m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");

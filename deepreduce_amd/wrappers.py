@@ -227,10 +227,13 @@ class DeepReduce(_WrapperBase):
             if self.params.get("pack_mapping", True):
                 # ceil(log2 k) bits per mapping entry instead of int32
                 # (paper App. E; the reference left this commented out at
-                # pytorch/deepreduce.py:264-265 — we ship it, GPU-packed)
+                # pytorch/deepreduce.py:264-265 — we ship it, GPU-packed).
+                # nbits is host-known (mapping is a permutation of
+                # arange(k)) — no device sync.
                 from .codecs.intpack import pack_with_header
 
-                mapping = pack_with_header(mapping.long())
+                nbits = max(1, (int(mapping.numel()) - 1).bit_length())
+                mapping = pack_with_header(mapping.long(), nbits=nbits)
                 tensors = (vals, idxs, mapping)
             else:
                 tensors = (vals, idxs, mapping.int())

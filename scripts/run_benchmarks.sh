@@ -26,7 +26,7 @@ case "${1:-all}" in
   config1)  # ResNet-20/CIFAR-10 top-k 1% + residual, gloo CPU ws=2 (no GPU)
     python -m pytest tests/test_distributed.py -q -k resnet20 ;;
   config2)  # ResNet-50 top-k 1% + Bloom index, N GPUs
-    launch "$NGPUS" --model resnet50 --deepreduce index --index bloom --overlap ;;
+    launch "$NGPUS" --model resnet50 --deepreduce index --index bloom ;;
   config3)  # ResNet-50 top-k 1% + polyfit value compression
     launch "$NGPUS" --model resnet50 --deepreduce value --value polyfit ;;
   config4)  # NCF embedding-heavy sparse grads
@@ -37,7 +37,7 @@ case "${1:-all}" in
   dense)    # uncompressed RCCL all-reduce baseline (the 1.0 of rel. volume)
     launch "$NGPUS" --model resnet50 --deepreduce dense ;;
   scaling)  # 1/2/4/8 scaling curve for config 2
-    for n in 1 2 4 8; do launch "$n" --model resnet50 --deepreduce index --overlap; done ;;
+    for n in 1 2 4 8; do launch "$n" --model resnet50 --deepreduce index; done ;;
   all)
     for c in config2 config3 config4 config5 dense; do "$0" "$c"; done ;;
   *) echo "usage: $0 {config1..config5|dense|scaling|all}"; exit 1 ;;

@@ -69,8 +69,25 @@ def _norm_x(segments, device):
     return seg_id, pos / seg_len[seg_id]
 
 
+def _seg_starts(segments, device):
+    seg_len = torch.as_tensor([0] + list(segments), dtype=torch.int64, device=device)
+    return torch.cumsum(seg_len, 0)
+
+
 def _fit_segments(y: torch.Tensor, segments, degree: int) -> torch.Tensor:
-    """[S, degree+1] float64 coefficients, batched normal equations."""
+    """[S, degree+1] float64 coefficients, batched normal equations.
+
+    GPU: ONE fused kernel (block per segment: moments + ridged Gram +
+    in-register Cholesky; ops/src/hip_ops.hip polyfit_fit_kernel) — the
+    torch index_add_ reduction below costs ~760 us per call in fp64
+    atomics.  CPU: the torch path.
+    """
+    from .. import ops
+
+    if y.is_cuda and ops.hip_available():
+        from deepreduce_amd import _hip_ops
+
+        return _hip_ops.polyfit_fit(y.float(), _seg_starts(segments, y.device), degree)
     device = y.device
     d1 = degree + 1
     seg_id, x = _norm_x(segments, device)
@@ -106,6 +123,13 @@ def _fit_segments(y: torch.Tensor, segments, degree: int) -> torch.Tensor:
 
 def _eval_segments(coeffs: torch.Tensor, segments, device) -> torch.Tensor:
     """Batched per-element polynomial evaluation (Horner, normalized x)."""
+    from .. import ops
+
+    if coeffs.is_cuda and ops.hip_available():
+        from deepreduce_amd import _hip_ops
+
+        N = int(sum(segments))
+        return _hip_ops.polyfit_eval(coeffs, _seg_starts(segments, coeffs.device), N).double()
     seg_id, x = _norm_x(segments, device)
     c = coeffs[seg_id]  # [N, d1]
     y = c[:, -1]

@@ -1457,6 +1457,159 @@ torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
     return dense;
 }
 
+
+// ---------------------------------------------------------------------------
+// Fused piecewise-polynomial fit (PolyFit/PolySeg value codecs).
+//
+// The torch path built an [N, 2d+1] float64 power matrix and reduced it
+// with index_add_ — fp64 atomics onto ~20 rows measured 760 us PER CALL
+// (profiles/NOTES.md r05).  Here: one block per segment accumulates the
+// power sums S_p = sum x^p and moments M_p = sum x^p*y (x = (i+1)/len,
+// the normalized abscissa) in registers, reduces through LDS, builds the
+// ridged Gram system and solves it in-register (Cholesky, d1 <= 8) —
+// one kernel for the whole fit, one for the eval.
+// ---------------------------------------------------------------------------
+
+#define PF_MAXD1 8  // degree <= 7
+
+__global__ void polyfit_fit_kernel(const float* __restrict__ y,
+                                   const int64_t* __restrict__ seg_starts /*[S+1]*/,
+                                   int degree, double* __restrict__ coeffs /*[S,d1]*/) {
+    const int s = blockIdx.x;
+    const int64_t start = seg_starts[s];
+    const int64_t end = seg_starts[s + 1];
+    const int64_t len = end - start;
+    const int d1 = degree + 1;
+    const int np = 2 * degree + 1;
+    const double inv_len = len > 0 ? 1.0 / (double)len : 1.0;
+
+    double ps[2 * PF_MAXD1 - 1];
+    double mo[PF_MAXD1];
+    for (int p = 0; p < np; ++p) ps[p] = 0.0;
+    for (int p = 0; p < d1; ++p) mo[p] = 0.0;
+
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+        const double x = (double)(i - start + 1) * inv_len;
+        const double yv = (double)y[i];
+        double xp = 1.0;
+        for (int p = 0; p < np; ++p) {
+            ps[p] += xp;
+            if (p < d1) mo[p] += xp * yv;
+            xp *= x;
+        }
+    }
+    // wave reduce, then cross-wave through LDS
+    const int lane = threadIdx.x % WAVE, wid = threadIdx.x / WAVE;
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        for (int p = 0; p < np; ++p) ps[p] += __shfl_down(ps[p], off, WAVE);
+        for (int p = 0; p < d1; ++p) mo[p] += __shfl_down(mo[p], off, WAVE);
+    }
+    __shared__ double sh[QBLOCK / WAVE][2 * PF_MAXD1 - 1 + PF_MAXD1];
+    if (lane == 0) {
+        for (int p = 0; p < np; ++p) sh[wid][p] = ps[p];
+        for (int p = 0; p < d1; ++p) sh[wid][np + p] = mo[p];
+    }
+    __syncthreads();
+    if (threadIdx.x != 0) return;
+    for (int w = 1; w < (int)(blockDim.x / WAVE); ++w) {
+        for (int p = 0; p < np; ++p) ps[p] += sh[w][p];
+        for (int p = 0; p < d1; ++p) mo[p] += sh[w][np + p];
+    }
+    // gram[i][j] = S_{i+j}; ridge = max|diag| * 1e-10 + 1e-30
+    double G[PF_MAXD1][PF_MAXD1];
+    double dmax = 0.0;
+    for (int i = 0; i < d1; ++i) {
+        double dg = fabs(ps[2 * i]);
+        if (dg > dmax) dmax = dg;
+    }
+    const double ridge = dmax * 1e-10 + 1e-30;
+    for (int i = 0; i < d1; ++i)
+        for (int j = 0; j < d1; ++j)
+            G[i][j] = ps[i + j] + (i == j ? ridge : 0.0);
+    // in-register Cholesky solve G c = mo
+    double L[PF_MAXD1][PF_MAXD1];
+    for (int i = 0; i < d1; ++i) {
+        for (int j = 0; j <= i; ++j) {
+            double sum = G[i][j];
+            for (int p = 0; p < j; ++p) sum -= L[i][p] * L[j][p];
+            if (i == j)
+                L[i][j] = sqrt(sum > 1e-300 ? sum : 1e-300);
+            else
+                L[i][j] = sum / L[j][j];
+        }
+    }
+    double yv[PF_MAXD1];
+    for (int i = 0; i < d1; ++i) {
+        double sum = mo[i];
+        for (int p = 0; p < i; ++p) sum -= L[i][p] * yv[p];
+        yv[i] = sum / L[i][i];
+    }
+    double* c = coeffs + (int64_t)s * d1;
+    for (int i = d1 - 1; i >= 0; --i) {
+        double sum = yv[i];
+        for (int p = i + 1; p < d1; ++p) sum -= L[p][i] * c[p];
+        c[i] = sum / L[i][i];
+    }
+}
+
+// Horner evaluation of the fitted piecewise polynomial (binary search for
+// the segment; S is tiny so the search stays in cache)
+__global__ void polyfit_eval_kernel(const double* __restrict__ coeffs,
+                                    const int64_t* __restrict__ seg_starts,
+                                    int S, int degree, int64_t N,
+                                    float* __restrict__ out) {
+    const int d1 = degree + 1;
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < N; i += stride) {
+        int lo = 0, hi = S - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (seg_starts[mid] <= i) lo = mid; else hi = mid - 1;
+        }
+        const int64_t start = seg_starts[lo];
+        const int64_t len = seg_starts[lo + 1] - start;
+        const double x = (double)(i - start + 1) / (double)(len > 0 ? len : 1);
+        const double* c = coeffs + (int64_t)lo * d1;
+        double yv = c[d1 - 1];
+        for (int p = d1 - 2; p >= 0; --p) yv = yv * x + c[p];
+        out[i] = (float)yv;
+    }
+}
+
+torch::Tensor polyfit_fit(torch::Tensor y, torch::Tensor seg_starts, int64_t degree) {
+    CHECK_CUDA(y);
+    TORCH_CHECK(degree + 1 <= PF_MAXD1, "degree <= 7");
+    auto yy = y.to(torch::kFloat32).contiguous();
+    auto ss = seg_starts.to(torch::kInt64).contiguous();
+    const int S = (int)ss.numel() - 1;
+    auto coeffs = torch::empty({S, degree + 1},
+                               torch::dtype(torch::kFloat64).device(y.device()));
+    if (S > 0) {
+        hipStream_t stream = at::hip::getCurrentHIPStream();
+        hipLaunchKernelGGL(polyfit_fit_kernel, dim3(S), dim3(QBLOCK), 0, stream,
+                           yy.data_ptr<float>(), ss.data_ptr<int64_t>(), (int)degree,
+                           coeffs.data_ptr<double>());
+    }
+    return coeffs;
+}
+
+torch::Tensor polyfit_eval(torch::Tensor coeffs, torch::Tensor seg_starts, int64_t N) {
+    CHECK_CUDA(coeffs);
+    auto cc = coeffs.to(torch::kFloat64).contiguous();
+    auto ss = seg_starts.to(torch::kInt64).contiguous();
+    const int S = (int)ss.numel() - 1;
+    const int degree = (int)cc.size(1) - 1;
+    auto out = torch::empty({N}, torch::dtype(torch::kFloat32).device(coeffs.device()));
+    if (N > 0) {
+        hipStream_t stream = at::hip::getCurrentHIPStream();
+        hipLaunchKernelGGL(polyfit_eval_kernel, dim3(bt_grid(N)), dim3(256), 0, stream,
+                           cc.data_ptr<double>(), ss.data_ptr<int64_t>(), S, degree, N,
+                           out.data_ptr<float>());
+    }
+    return out;
+}
+
 // ---------------------------------------------------------------------------
 // CPU-native C++ paths (replace the reference's TF C++ CPU ops:
 // bloom_filter_compression.cc / integer_compression.cc) — same wire format
@@ -1586,6 +1739,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "own-payload decode: (wire, out_idx) -> dense flat");
     m.def("batched_decode_sum", &batched_decode_sum,
           "multi-rank decode: [R, W] wires -> sum of dense decodes");
+    m.def("polyfit_fit", &polyfit_fit,
+          "fused piecewise polynomial fit: moments+gram+cholesky per segment");
+    m.def("polyfit_eval", &polyfit_eval, "fused piecewise Horner eval");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
     m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");

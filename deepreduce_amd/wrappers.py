@@ -219,10 +219,13 @@ class DeepReduce(_WrapperBase):
             vals, idxs, _ = self.idx_codec.compress((vals, idxs, tensor.size()), self.params)
             self.params.pop("dense_tensor", None)
             own = self.params.pop("_own_decoded", None)
-            if own is not None:
-                self._own_cache[name] = own[1]  # bloom-recovered indices
             new_idxs = torch.arange(vals.numel(), device=vals.device)
             vals, mapping, shape_out = self.val_codec.compress((vals, new_idxs, shape), self.params)
+            if own is not None:
+                # cache (bloom indices, unpacked mapping): decompress_own
+                # then skips both the universe query AND the mapping
+                # header parse (a host sync per tensor)
+                self._own_cache[name] = (own[1], mapping.long())
             ctx = shape_out
             if self.params.get("pack_mapping", True):
                 # ceil(log2 k) bits per mapping entry instead of int32
@@ -268,14 +271,15 @@ class DeepReduce(_WrapperBase):
         return mapping.long()
 
     def decompress_own(self, tensors, ctx, name):
-        """Own-payload decode with the cached bloom indices: only the (cheap)
-        value-codec eval runs; the full-universe query is skipped."""
-        cached_idxs = self._own_cache.get(name)
+        """Own-payload decode with the cached bloom indices and mapping:
+        only the (cheap) value-codec eval runs; the full-universe query
+        and the mapping header parse are both skipped."""
+        cached = self._own_cache.get(name)
         shape = ctx
-        if cached_idxs is None or torch.Size(shape).numel() <= _BYPASS_NUMEL:
+        if cached is None or torch.Size(shape).numel() <= _BYPASS_NUMEL:
             return self.decompress(tensors, ctx)
-        vals, idxs, mapping = tensors
-        mapping = self._unpack_mapping(mapping)
+        cached_idxs, mapping = cached
+        vals = tensors[0]
         vals, _, _ = self.val_codec.decompress((vals, mapping, shape), self.params)
         idxs = cached_idxs[mapping]
         return self.sparsifier.decompress((vals, idxs), shape)

@@ -352,3 +352,21 @@ def test_batched_multi_rank_decode(dev):
         off_bytes += pad8(4 * k) + pad8(nb)
         off_vals += n
     torch.cuda.synchronize()
+
+
+def test_polyfit_fused_kernel_parity(hip, dev):
+    """Fused fit/eval kernels vs the torch float64 reference path."""
+    from deepreduce_amd.codecs import polyfit as pf
+
+    torch.manual_seed(3)
+    y = torch.sort(torch.randn(20_000).abs(), descending=True).values
+    segments = pf.get_segments(20_000, 20_000)  # all-positive curve
+    deg = 5
+    ref = pf._fit_segments(y.double(), segments, deg)          # CPU torch path
+    got = hip.polyfit_fit(y.to(dev), pf._seg_starts(segments, dev), deg).cpu()
+    assert torch.allclose(ref, got, rtol=1e-6, atol=1e-9), (ref - got).abs().max()
+
+    ev_ref = pf._eval_segments(ref, segments, torch.device("cpu")).float()
+    ev_got = hip.polyfit_eval(got.to(dev), pf._seg_starts(segments, dev), y.numel()).cpu()
+    assert torch.allclose(ev_ref, ev_got, rtol=1e-5, atol=1e-6), \
+        (ev_ref - ev_got).abs().max()

@@ -27,19 +27,20 @@ from . import SparseCompressor
 _RATIOS = [1 / 5, 1 / 10, 1 / 30, 1 / 100, 1 / 300, 1 / 1000, 1 / 3000, 1 / 10000, 1 / 30000, 1 / 100000]
 
 
-def get_segments(N: int, num_pos: int = 0):
-    """Geometric segmentation split at the pos/neg boundary.
+S_PAD = 2 * len(_RATIOS) + 2  # fixed 22-slot segment layout
 
-    Exact integer-math parity with pytorch/deepreduce.py:362-377 — this runs
-    identically on every rank from (N, num_pos) alone.
+
+def get_segments(N: int, num_pos: int = 0):
+    """Geometric segmentation split at the pos/neg boundary, PADDED to a
+    fixed 22-slot layout (zero-length slots where the reference drops the
+    segment, pytorch/deepreduce.py:362-377 — per-slot boundary math is
+    identical).  The padding makes the payload size a function of N alone:
+    uniform allgather payloads, and the GPU path derives the boundaries
+    on-device from the transmitted num_pos with NO host sync.
     """
-    pos, neg = [], []
     num_neg = N - num_pos
-    for r in _RATIOS:
-        if int(num_pos * r) > 30:
-            pos.append(int(num_pos * r))
-        if int(num_neg * r) > 30:
-            neg.append(int(num_neg * r))
+    pos = [int(num_pos * r) if int(num_pos * r) > 30 else 0 for r in _RATIOS]
+    neg = [int(num_neg * r) if int(num_neg * r) > 30 else 0 for r in _RATIOS]
     return pos[::-1] + [num_pos - sum(pos)] + [num_neg - sum(neg)] + neg
 
 
@@ -143,19 +144,32 @@ class PolyFit(SparseCompressor):
 
     @staticmethod
     def compress(sparse_tensor, params):
+        from .. import ops
+
         degree = int(params.get("poly_degree", 5))
         sort = params.get("sort", False)
         vals, idxs, shape = sparse_tensor
         N = int(idxs.numel())
         y = vals.float()
-        num_pos = int((y > 0).sum().item())
 
         if not sort:  # values arrive unsorted: sort desc, remember mapping
             y, mapping = y.sort(descending=True)
             idxs = idxs[mapping]
 
+        if y.is_cuda and ops.hip_available():
+            # fully sync-free: num_pos stays on device; segment starts and
+            # the fused fit run in two kernels
+            from deepreduce_amd import _hip_ops
+
+            num_pos_t = (y > 0).sum().double()
+            starts = _hip_ops.polyfit_starts(num_pos_t, N)
+            coeffs = _hip_ops.polyfit_fit(y, starts, degree)
+            payload = torch.cat([coeffs.reshape(-1), num_pos_t.reshape(1)])
+            return payload, idxs, shape
+
+        num_pos = int((y > 0).sum().item())
         segments = get_segments(N, num_pos)
-        coeffs = _fit_segments(y, segments, degree)  # [S, d1]
+        coeffs = _fit_segments(y, segments, degree)  # [S_PAD, d1]
         payload = torch.cat(
             [coeffs.reshape(-1), torch.tensor([float(num_pos)], dtype=torch.float64, device=y.device)]
         )
@@ -163,12 +177,20 @@ class PolyFit(SparseCompressor):
 
     @staticmethod
     def decompress(fitted_sparse_tensor, params):
+        from .. import ops
+
         payload, idxs, shape = fitted_sparse_tensor
         N = int(idxs.numel())
         coeffs_flat, num_pos_t = payload.split([payload.numel() - 1, 1])
+        d1 = coeffs_flat.numel() // S_PAD
+        if payload.is_cuda and ops.hip_available():
+            from deepreduce_amd import _hip_ops
+
+            starts = _hip_ops.polyfit_starts(num_pos_t.double(), N)
+            vals = _hip_ops.polyfit_eval(coeffs_flat.reshape(S_PAD, d1), starts, N)
+            return vals, idxs, shape
         num_pos = int(num_pos_t.item())
         segments = get_segments(N, num_pos)
-        d1 = coeffs_flat.numel() // len(segments)
-        coeffs = coeffs_flat.reshape(len(segments), d1)
+        coeffs = coeffs_flat.reshape(S_PAD, d1)
         vals = _eval_segments(coeffs, segments, payload.device).float()
         return vals, idxs, shape

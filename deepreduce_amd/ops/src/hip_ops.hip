@@ -1577,6 +1577,52 @@ __global__ void polyfit_eval_kernel(const double* __restrict__ coeffs,
     }
 }
 
+
+// Segment boundaries of the padded 22-slot polyfit layout, derived
+// ON-DEVICE from num_pos (codecs/polyfit.py get_segments math) — keeps the
+// whole codec sync-free.  ratios = 1/5 .. 1/100000; a slot is zero-length
+// when int(num*r) <= 30 (reference condition, pytorch/deepreduce.py:370-374).
+__global__ void polyfit_starts_kernel(const double* __restrict__ num_pos_p, int64_t N,
+                                      int64_t* __restrict__ starts /*[2*NR+3]*/) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    const double RA[10] = {1.0 / 5, 1.0 / 10, 1.0 / 30, 1.0 / 100, 1.0 / 300,
+                           1.0 / 1000, 1.0 / 3000, 1.0 / 10000, 1.0 / 30000,
+                           1.0 / 100000};
+    const int NR = 10;
+    int64_t np = (int64_t)(*num_pos_p + 0.5);
+    int64_t nn = N - np;
+    int64_t pos[10], neg[10], psum = 0, nsum = 0;
+    for (int i = 0; i < NR; ++i) {
+        int64_t p = (int64_t)((double)np * RA[i]);
+        int64_t n = (int64_t)((double)nn * RA[i]);
+        pos[i] = (p > 30) ? p : 0;
+        neg[i] = (n > 30) ? n : 0;
+        psum += pos[i];
+        nsum += neg[i];
+    }
+    int64_t seg[2 * 10 + 2];
+    for (int i = 0; i < NR; ++i) seg[i] = pos[NR - 1 - i];  // reversed
+    seg[NR] = np - psum;
+    seg[NR + 1] = nn - nsum;
+    for (int i = 0; i < NR; ++i) seg[NR + 2 + i] = neg[i];
+    int64_t acc = 0;
+    starts[0] = 0;
+    for (int i = 0; i < 2 * NR + 2; ++i) {
+        acc += seg[i];
+        starts[i + 1] = acc;
+    }
+}
+
+torch::Tensor polyfit_starts(torch::Tensor num_pos, int64_t N) {
+    CHECK_CUDA(num_pos);
+    auto npd = num_pos.to(torch::kFloat64).contiguous();
+    auto starts = torch::empty({23}, torch::dtype(torch::kInt64).device(num_pos.device()));
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(polyfit_starts_kernel, dim3(1), dim3(WAVE), 0, stream,
+                       npd.data_ptr<double>(), N, starts.data_ptr<int64_t>());
+    return starts;
+}
+
 torch::Tensor polyfit_fit(torch::Tensor y, torch::Tensor seg_starts, int64_t degree) {
     CHECK_CUDA(y);
     TORCH_CHECK(degree + 1 <= PF_MAXD1, "degree <= 7");
@@ -1742,6 +1788,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("polyfit_fit", &polyfit_fit,
           "fused piecewise polynomial fit: moments+gram+cholesky per segment");
     m.def("polyfit_eval", &polyfit_eval, "fused piecewise Horner eval");
+    m.def("polyfit_starts", &polyfit_starts,
+          "device-side padded segment boundaries from num_pos");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
     m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");

@@ -72,9 +72,9 @@ class ValueCompressor(_WrapperBase):
         super().__init__(sparsifier, params)
         name = self.params.get("value", "polyfit")
         self.val_codec = codec_registry[name]
-        if name not in ("qsgd", "polyseg"):
-            # polyfit coefficient count varies with per-rank num_pos;
-            # qsgd and polyseg payload sizes depend only on k -> uniform
+        if name not in ("qsgd", "polyseg", "polyfit"):
+            # qsgd/polyseg payload sizes depend only on k; polyfit's padded
+            # 22-slot layout depends only on N -> all uniform across ranks
             self.tensors_size_are_same = False
 
     def compress(self, tensor, name):
@@ -200,9 +200,19 @@ class DeepReduce(_WrapperBase):
 
     def __init__(self, sparsifier, params=None):
         super().__init__(sparsifier, params)
-        self.val_codec = codec_registry[self.params.get("value", "polyfit")]
-        self.idx_codec = codec_registry[self.params.get("index", "bloom")]
-        self.tensors_size_are_same = False  # coeff counts vary with num_pos
+        value = self.params.get("value", "polyfit")
+        index = self.params.get("index", "bloom")
+        self.val_codec = codec_registry[value]
+        self.idx_codec = codec_registry[index]
+        # polyfit (padded layout), qsgd, polyseg payloads depend only on
+        # (N, k); bloom bits + bit-packed mapping likewise -> uniform when
+        # the sparsifier is uniform and the policy is not P0
+        self.tensors_size_are_same = (
+            sparsifier.tensors_size_are_same
+            and value in ("polyfit", "qsgd", "polyseg")
+            and index == "bloom"
+            and self.params.get("policy", "leftmost") != "p0"
+        )
 
     def compress(self, tensor, name):
         tensors, ctx = self.sparsifier.compress(tensor, name)

@@ -370,3 +370,41 @@ def test_polyfit_fused_kernel_parity(hip, dev):
     ev_got = hip.polyfit_eval(got.to(dev), pf._seg_starts(segments, dev), y.numel()).cpu()
     assert torch.allclose(ev_ref, ev_got, rtol=1e-5, atol=1e-6), \
         (ev_ref - ev_got).abs().max()
+
+
+def test_polyfit_starts_kernel_parity(hip, dev):
+    """Device-side padded segment boundaries == python get_segments."""
+    from deepreduce_amd.codecs.polyfit import get_segments
+
+    for N, num_pos in [(20_000, 7_345), (1_500, 0), (1_500, 1_500),
+                       (2_000_000, 999_999), (1_001, 500)]:
+        seg = get_segments(N, num_pos)
+        expect = [0]
+        for s in seg:
+            expect.append(expect[-1] + s)
+        npt = torch.tensor([float(num_pos)], dtype=torch.float64, device=dev)
+        got = hip.polyfit_starts(npt, N).cpu().tolist()
+        assert got == expect, (N, num_pos)
+
+
+def test_polyfit_gpu_sync_free_roundtrip(dev):
+    """GPU polyfit compress/decompress round trip via the codec API; payload
+    size must be uniform (function of N only)."""
+    from deepreduce_amd.codecs import compressor
+
+    torch.manual_seed(21)
+    N = 30_000
+    sizes = set()
+    for trial in range(3):
+        vals = torch.randn(N, device=dev) * (trial + 1)
+        idxs = torch.randperm(3_000_000, device=dev)[:N]
+        p, m, shape = compressor["polyfit"].compress(
+            (vals, idxs, torch.Size([3_000_000])), {"poly_degree": 5})
+        sizes.add(p.numel())
+        v2, i2, _ = compressor["polyfit"].decompress((p, m, shape), {"poly_degree": 5})
+        assert v2.numel() == N
+        # fit quality: relative L2 error of the sorted curve reconstruction
+        ref = vals.sort(descending=True).values
+        err = (v2 - ref).norm() / ref.norm()
+        assert err < 0.15, float(err)
+    assert len(sizes) == 1  # uniform payload size

@@ -81,7 +81,8 @@ def _graph_safe(grc) -> bool:
     policy = params.get("policy", "leftmost")
     if mode in ("index", "both") and (index != "bloom" or policy != "leftmost"):
         return False
-    if mode in ("value", "both") and value not in ("qsgd",):
+    if mode in ("value", "both") and value not in ("qsgd", "polyfit", "polyseg"):
+        # polyfit is sync-free on GPU (device-derived padded segments)
         return False
     if params.get("micro-benchmark"):
         return False  # timing prints sync

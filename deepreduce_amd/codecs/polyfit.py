@@ -27,20 +27,30 @@ from . import SparseCompressor
 _RATIOS = [1 / 5, 1 / 10, 1 / 30, 1 / 100, 1 / 300, 1 / 1000, 1 / 3000, 1 / 10000, 1 / 30000, 1 / 100000]
 
 
-S_PAD = 2 * len(_RATIOS) + 2  # fixed 22-slot segment layout
+def active_ratios(N: int):
+    """Ratios whose segment CAN be nonzero for any num_pos <= N
+    (int(N*r) > 30 — the reference's inclusion condition applied to the
+    upper bound).  Host-known from N alone, so identical on every rank."""
+    return [r for r in _RATIOS if int(N * r) > 30]
+
+
+def s_pad(N: int) -> int:
+    """Padded slot count: payload size is a function of N alone."""
+    return 2 * len(active_ratios(N)) + 2
 
 
 def get_segments(N: int, num_pos: int = 0):
-    """Geometric segmentation split at the pos/neg boundary, PADDED to a
-    fixed 22-slot layout (zero-length slots where the reference drops the
-    segment, pytorch/deepreduce.py:362-377 — per-slot boundary math is
-    identical).  The padding makes the payload size a function of N alone:
-    uniform allgather payloads, and the GPU path derives the boundaries
-    on-device from the transmitted num_pos with NO host sync.
+    """Geometric segmentation split at the pos/neg boundary, PADDED to the
+    fixed s_pad(N)-slot layout (zero-length slots where the reference
+    drops the segment, pytorch/deepreduce.py:362-377 — per-slot boundary
+    math is identical).  The padding makes the payload size a function of
+    N alone: uniform allgather payloads, and the GPU path derives the
+    boundaries on-device from the transmitted num_pos with NO host sync.
     """
     num_neg = N - num_pos
-    pos = [int(num_pos * r) if int(num_pos * r) > 30 else 0 for r in _RATIOS]
-    neg = [int(num_neg * r) if int(num_neg * r) > 30 else 0 for r in _RATIOS]
+    ratios = active_ratios(N)
+    pos = [int(num_pos * r) if int(num_pos * r) > 30 else 0 for r in ratios]
+    neg = [int(num_neg * r) if int(num_neg * r) > 30 else 0 for r in ratios]
     return pos[::-1] + [num_pos - sum(pos)] + [num_neg - sum(neg)] + neg
 
 
@@ -169,7 +179,7 @@ class PolyFit(SparseCompressor):
 
         num_pos = int((y > 0).sum().item())
         segments = get_segments(N, num_pos)
-        coeffs = _fit_segments(y, segments, degree)  # [S_PAD, d1]
+        coeffs = _fit_segments(y, segments, degree)  # [s_pad(N), d1]
         payload = torch.cat(
             [coeffs.reshape(-1), torch.tensor([float(num_pos)], dtype=torch.float64, device=y.device)]
         )
@@ -182,15 +192,16 @@ class PolyFit(SparseCompressor):
         payload, idxs, shape = fitted_sparse_tensor
         N = int(idxs.numel())
         coeffs_flat, num_pos_t = payload.split([payload.numel() - 1, 1])
-        d1 = coeffs_flat.numel() // S_PAD
+        sp = s_pad(N)
+        d1 = coeffs_flat.numel() // sp
         if payload.is_cuda and ops.hip_available():
             from deepreduce_amd import _hip_ops
 
             starts = _hip_ops.polyfit_starts(num_pos_t.double(), N)
-            vals = _hip_ops.polyfit_eval(coeffs_flat.reshape(S_PAD, d1), starts, N)
+            vals = _hip_ops.polyfit_eval(coeffs_flat.reshape(sp, d1), starts, N)
             return vals, idxs, shape
         num_pos = int(num_pos_t.item())
         segments = get_segments(N, num_pos)
-        coeffs = coeffs_flat.reshape(S_PAD, d1)
+        coeffs = coeffs_flat.reshape(sp, d1)
         vals = _eval_segments(coeffs, segments, payload.device).float()
         return vals, idxs, shape

@@ -1591,8 +1591,13 @@ __global__ void polyfit_starts_kernel(const double* __restrict__ num_pos_p, int6
     const int NR = 10;
     int64_t np = (int64_t)(*num_pos_p + 0.5);
     int64_t nn = N - np;
+    // active ratio subset: int(N*r) > 30 (host derives the same count, so
+    // the payload slot layout matches codecs/polyfit.py s_pad(N))
+    int NA = 0;
+    for (int i = 0; i < NR; ++i)
+        if ((int64_t)((double)N * RA[i]) > 30) NA = i + 1;
     int64_t pos[10], neg[10], psum = 0, nsum = 0;
-    for (int i = 0; i < NR; ++i) {
+    for (int i = 0; i < NA; ++i) {
         int64_t p = (int64_t)((double)np * RA[i]);
         int64_t n = (int64_t)((double)nn * RA[i]);
         pos[i] = (p > 30) ? p : 0;
@@ -1601,22 +1606,33 @@ __global__ void polyfit_starts_kernel(const double* __restrict__ num_pos_p, int6
         nsum += neg[i];
     }
     int64_t seg[2 * 10 + 2];
-    for (int i = 0; i < NR; ++i) seg[i] = pos[NR - 1 - i];  // reversed
-    seg[NR] = np - psum;
-    seg[NR + 1] = nn - nsum;
-    for (int i = 0; i < NR; ++i) seg[NR + 2 + i] = neg[i];
+    for (int i = 0; i < NA; ++i) seg[i] = pos[NA - 1 - i];  // reversed
+    seg[NA] = np - psum;
+    seg[NA + 1] = nn - nsum;
+    for (int i = 0; i < NA; ++i) seg[NA + 2 + i] = neg[i];
     int64_t acc = 0;
     starts[0] = 0;
-    for (int i = 0; i < 2 * NR + 2; ++i) {
+    for (int i = 0; i < 2 * NA + 2; ++i) {
         acc += seg[i];
         starts[i + 1] = acc;
     }
 }
 
+static inline int64_t polyfit_s_pad(int64_t N) {
+    const double RA[10] = {1.0 / 5, 1.0 / 10, 1.0 / 30, 1.0 / 100, 1.0 / 300,
+                           1.0 / 1000, 1.0 / 3000, 1.0 / 10000, 1.0 / 30000,
+                           1.0 / 100000};
+    int64_t na = 0;
+    for (int i = 0; i < 10; ++i)
+        if ((int64_t)((double)N * RA[i]) > 30) na = i + 1;
+    return 2 * na + 2;
+}
+
 torch::Tensor polyfit_starts(torch::Tensor num_pos, int64_t N) {
     CHECK_CUDA(num_pos);
     auto npd = num_pos.to(torch::kFloat64).contiguous();
-    auto starts = torch::empty({23}, torch::dtype(torch::kInt64).device(num_pos.device()));
+    auto starts = torch::empty({polyfit_s_pad(N) + 1},
+                               torch::dtype(torch::kInt64).device(num_pos.device()));
     hipStream_t stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(polyfit_starts_kernel, dim3(1), dim3(WAVE), 0, stream,
                        npd.data_ptr<double>(), N, starts.data_ptr<int64_t>());

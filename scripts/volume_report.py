@@ -62,7 +62,11 @@ def measure(ratio: float, steps: int):
     rows = []
     for label, extra in CONFIGS:
         params = {"compressor": "topk", "memory": "residual",
-                  "communicator": "allgather", "compress_ratio": ratio}
+                  "communicator": "allgather", "compress_ratio": ratio,
+                  # codec-pure accounting: the runtime's small-tensor dense
+                  # fusion (a latency optimization) is off here so the
+                  # table isolates what each CODEC transmits
+                  "small_dense": False}
         params.update(extra)
         grc = deepreduce_from_params(params)
         total = 0

@@ -408,3 +408,40 @@ def test_polyfit_gpu_sync_free_roundtrip(dev):
         err = (v2 - ref).norm() / ref.norm()
         assert err < 0.15, float(err)
     assert len(sizes) == 1  # uniform payload size
+
+
+def test_graph_vs_eager_both_mode(dev):
+    """hipGraph replay of the full 'both' (bloom+polyfit) pipeline must
+    match eager training step-for-step."""
+    from deepreduce_amd import DistributedOptimizer, deepreduce_from_params
+
+    def run(use_graph):
+        torch.manual_seed(17)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(256, 384), torch.nn.ReLU(), torch.nn.Linear(384, 10)
+        ).to(dev)
+        params = {
+            "compressor": "topk", "memory": "residual",
+            "communicator": "allgather", "compress_ratio": 0.02,
+            "deepreduce": "both", "index": "bloom", "policy": "leftmost",
+            "value": "polyfit",
+        }
+        grc = deepreduce_from_params(params)
+        opt = DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.05), grc, model,
+            use_graph=use_graph, graph_warmup=2,
+        )
+        gen = torch.Generator(device="cpu").manual_seed(23)
+        for _ in range(7):
+            x = torch.randn(64, 256, generator=gen).to(dev)
+            y = torch.randint(0, 10, (64,), generator=gen).to(dev)
+            opt.zero_grad(set_to_none=False)
+            torch.nn.functional.cross_entropy(model(x), y).backward()
+            opt.step()
+        torch.cuda.synchronize()
+        return [p.detach().clone() for p in model.parameters()]
+
+    eager = run(False)
+    graphed = run(True)
+    for a, b in zip(eager, graphed):
+        assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()

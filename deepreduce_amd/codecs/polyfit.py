@@ -163,7 +163,9 @@ class PolyFit(SparseCompressor):
         y = vals.float()
 
         if not sort:  # values arrive unsorted: sort desc, remember mapping
-            y, mapping = y.sort(descending=True)
+            # stable: deterministic tie order, bit-identical to the batched
+            # whole-model pipeline (ops/batched.py BothPipeline)
+            y, mapping = torch.sort(y, dim=0, descending=True, stable=True)
             idxs = idxs[mapping]
 
         if y.is_cuda and ops.hip_available():

@@ -251,9 +251,8 @@ class Allgather(Communicator):
         else:
             c_flat = torch.cat([t.reshape(-1) for t in compensated])
 
-        wire, out_idx = bp.compress(c_flat)
+        wire, own_dense = bp.compress_and_own(c_flat)
         self.last_wire_bytes = int(wire.numel())
-        own_dense = bp.decode_own(wire, out_idx)
 
         # residual <- compensated - own decode (exactly the generic math)
         if isinstance(mem, ResidualMemory):

@@ -26,6 +26,10 @@ def _pad8(x: int) -> int:
 
 
 class BatchedPipeline:
+    """Index-mode (topk + bloom leftmost) whole-model pipeline."""
+
+    kind = "index"
+
     def __init__(self, names, numels, params, device):
         from ..codecs.bloom import Bloom
 
@@ -34,7 +38,7 @@ class BatchedPipeline:
         T = len(numels)
         ratio = params.get("compress_ratio", 0.01)
 
-        desc = torch.zeros(T, 12, dtype=torch.int64)
+        desc = torch.zeros(T, 16, dtype=torch.int64)
         b2t = []
         voff = koff = wire_off = cntoff = mwoff = blkoff = 0
         self.metas = []
@@ -80,6 +84,10 @@ class BatchedPipeline:
         )
         return wire, out_idx
 
+    def compress_and_own(self, values_flat: torch.Tensor):
+        wire, out_idx = self.compress(values_flat)
+        return wire, self.decode_own(wire, out_idx)
+
     def decode_own(self, wire, out_idx):
         from deepreduce_amd import _hip_ops
 
@@ -93,17 +101,122 @@ class BatchedPipeline:
                                            self.total_values, self.mask_words)
 
 
+class BothPipeline(BatchedPipeline):
+    """'both'-mode whole-model pipeline: bloom index + polyfit values +
+    bit-packed mapping, wire-compatible with the generic DeepReduce
+    wrapper payload (coeffs f64 | bloom bits | 5B-header packed mapping).
+    """
+
+    kind = "both"
+
+    def __init__(self, names, numels, params, device):
+        from ..codecs.bloom import Bloom
+        from ..codecs.polyfit import s_pad
+
+        self.names = list(names)
+        self.numels = list(numels)
+        self.degree = int(params.get("poly_degree", 5))
+        d1 = self.degree + 1
+        T = len(numels)
+        ratio = params.get("compress_ratio", 0.01)
+
+        desc = torch.zeros(T, 16, dtype=torch.int64)
+        b2t, seg_t, seg_i = [], [], []
+        voff = koff = wire_off = cntoff = mwoff = blkoff = 0
+        kmax = 1
+        self.metas = []
+        for t, n in enumerate(numels):
+            k = max(1, int(round(n * ratio)))
+            kmax = max(kmax, k)
+            num_hash, m = Bloom._config(k, n, params)
+            nbytes = (m + 7) // 8
+            nb = (n + BT_CHUNK - 1) // BT_CHUNK
+            sp = s_pad(k)
+            nbits = max(1, (k - 1).bit_length())
+            coeff_bytes = (sp * d1 + 1) * 8
+            map_payload = 5 + (k * nbits + 7) // 8
+            desc[t, 0] = n
+            desc[t, 1] = voff
+            desc[t, 2] = k
+            desc[t, 3] = koff
+            desc[t, 4] = m
+            desc[t, 5] = num_hash
+            desc[t, 6] = wire_off + coeff_bytes          # bloom bits
+            desc[t, 8] = cntoff
+            desc[t, 9] = mwoff
+            desc[t, 10] = blkoff
+            desc[t, 11] = sp
+            desc[t, 12] = wire_off                       # coeffs
+            desc[t, 13] = wire_off + coeff_bytes + _pad8(nbytes)  # mapping
+            desc[t, 14] = nbits
+            self.metas.append([(torch.float64, sp * d1 + 1),
+                               (torch.uint8, nbytes),
+                               (torch.uint8, map_payload)])
+            b2t.extend([t] * nb)
+            seg_t.extend([t] * sp)
+            seg_i.extend(range(sp))
+            voff += n
+            koff += k
+            wire_off += coeff_bytes + _pad8(nbytes) + _pad8(map_payload)
+            cntoff += nb
+            mwoff += nb * (BT_CHUNK // 64)
+            blkoff += nb
+
+        self.total_values = voff
+        self.k_total = koff
+        self.kmax = kmax
+        self.wire_bytes = wire_off
+        self.mask_words = mwoff
+        self.desc = desc.to(device)
+        self.b2t = torch.tensor(b2t, dtype=torch.int32, device=device)
+        self.seg_t = torch.tensor(seg_t, dtype=torch.int32, device=device)
+        self.seg_i = torch.tensor(seg_i, dtype=torch.int32, device=device)
+
+    def compress_and_own(self, values_flat: torch.Tensor):
+        from deepreduce_amd import _hip_ops
+
+        wire, own = _hip_ops.batched_compress_both(
+            values_flat, self.desc, self.b2t, self.seg_t, self.seg_i,
+            self.wire_bytes, self.k_total, self.mask_words, self.kmax,
+            self.degree, self.total_values,
+        )
+        return wire, own
+
+    def decode_sum(self, wires2d):
+        from deepreduce_amd import _hip_ops
+
+        return _hip_ops.batched_decode_both_sum(
+            wires2d, self.desc, self.b2t, self.total_values, self.mask_words,
+            self.k_total, self.degree,
+        )
+
+
 def maybe_pipeline(communicator, comp, named_tensors):
-    """Return a cached BatchedPipeline when the configuration qualifies."""
+    """Return a cached pipeline when the configuration qualifies:
+    BatchedPipeline for IndexCompressor(topk+bloom+leftmost),
+    BothPipeline for DeepReduce(topk+bloom+leftmost+polyfit)."""
     from . import hip_available
     from ..codecs.bloom import Bloom
+    from ..codecs.polyfit import PolyFit
     from ..compressors import TopKCompressor
-    from ..wrappers import IndexCompressor
+    from ..wrappers import DeepReduce, IndexCompressor
 
-    if not isinstance(comp, IndexCompressor) or comp.idx_codec is not Bloom:
+    cls = None
+    if isinstance(comp, IndexCompressor) and comp.idx_codec is Bloom:
+        cls = BatchedPipeline
+    elif (isinstance(comp, DeepReduce) and comp.idx_codec is Bloom
+          and comp.val_codec is PolyFit):
+        cls = BothPipeline
+    if cls is None:
         return None
     params = comp.params
     if params.get("policy", "leftmost") != "leftmost" or params.get("micro-benchmark"):
+        return None
+    if cls is BothPipeline and (
+        not params.get("pack_mapping", True)
+        or not params.get("fp_aware", True)
+        or params.get("sort", False)
+    ):
         return None
     if not isinstance(comp.sparsifier, TopKCompressor):
         return None
@@ -117,16 +230,18 @@ def maybe_pipeline(communicator, comp, named_tensors):
     if communicator.world_size > 16:
         return None
     key = (
+        cls.__name__,
         tuple(n for n, _ in named_tensors),
         tuple(t.numel() for t in grads),
         comp.sparsifier.compress_ratio,
         params.get("fpr"),
+        params.get("poly_degree", 5),
         str(grads[0].device),
     )
     cached = getattr(communicator, "_bt_pipeline", None)
     if cached is not None and cached[0] == key:
         return cached[1]
-    bp = BatchedPipeline([n for n, _ in named_tensors],
-                         [t.numel() for t in grads], params, grads[0].device)
+    bp = cls([n for n, _ in named_tensors],
+             [t.numel() for t in grads], params, grads[0].device)
     communicator._bt_pipeline = (key, bp)
     return bp

@@ -885,7 +885,7 @@ torch::Tensor cholesky_solve_small(torch::Tensor G, torch::Tensor b) {
 // ===========================================================================
 
 #define BT_CHUNK 8192  // elements per block (multiple of 256)
-#define BT_F 12
+#define BT_F 16
 
 __device__ __forceinline__ const int64_t* bt_row(const int64_t* desc, int t) {
     return desc + (int64_t)t * BT_F;
@@ -1205,10 +1205,11 @@ __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
 // positions decompress will re-derive
 __global__ void bt_qscatter_own_kernel(const uint64_t* __restrict__ mask,
                                        const int* __restrict__ qoffs,
-                                       const float* __restrict__ vals,
+                                       const float* __restrict__ vals /*nullable*/,
                                        const int64_t* __restrict__ desc,
                                        const int* __restrict__ b2t,
-                                       uint8_t* __restrict__ wire,
+                                       uint8_t* __restrict__ wire /*nullable*/,
+                                       float* __restrict__ vals_out /*nullable*/,
                                        int64_t* __restrict__ out_idx) {
     const int t = b2t[blockIdx.x];
     const int64_t* D = bt_row(desc, t);
@@ -1217,8 +1218,9 @@ __global__ void bt_qscatter_own_kernel(const uint64_t* __restrict__ mask,
     const int64_t end = min(start + BT_CHUNK, D[0]);
     const int64_t k = D[2];
     const int64_t mwoff = D[9];
-    const float* __restrict__ v = vals + D[1];
-    float* __restrict__ wv = (float*)(wire + D[7]);
+    const float* __restrict__ v = vals ? vals + D[1] : nullptr;
+    float* __restrict__ wv = wire ? (float*)(wire + D[7]) : nullptr;
+    float* __restrict__ vo = vals_out ? vals_out + D[3] : nullptr;
     int64_t* __restrict__ oi = out_idx + D[3];
     int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
     __shared__ int wave_cnt[QBLOCK / WAVE];
@@ -1238,7 +1240,8 @@ __global__ void bt_qscatter_own_kernel(const uint64_t* __restrict__ mask,
             int64_t ord = base_s + wbase + __popcll(ball & below);
             if (ord < k) {
                 oi[ord] = i;
-                wv[ord] = v[i];
+                if (wv) wv[ord] = v[i];
+                if (vo) vo[ord] = v[i];
             }
         }
         __syncthreads();
@@ -1392,7 +1395,8 @@ std::vector<torch::Tensor> batched_compress(torch::Tensor values_flat,
                        qcounts, dp, T, BV, qoffs);
     hipLaunchKernelGGL(bt_qscatter_own_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
                        (const uint64_t*)mask.data_ptr<int64_t>(), qoffs, vp, dp, mp,
-                       wire.data_ptr<uint8_t>(), out_idx.data_ptr<int64_t>());
+                       wire.data_ptr<uint8_t>(), (float*)nullptr,
+                       out_idx.data_ptr<int64_t>());
     return {wire, out_idx};
 }
 
@@ -1672,6 +1676,489 @@ torch::Tensor polyfit_eval(torch::Tensor coeffs, torch::Tensor seg_starts, int64
     return out;
 }
 
+
+// ===========================================================================
+// Batched 'both'-mode pipeline (bloom index + polyfit values + packed
+// mapping), building on the bt_* machinery.  Wire layout per tensor —
+// byte-identical to the generic _flatten_payload of the 'both' wrapper:
+//   [coeffs f64 (sp*d1+1), pad8] [bloom bits u8 ceil(m/8), pad8]
+//   [mapping u8: 5B header (count LE32, nbits) + ceil(k*nbits/8), pad8]
+// Extra descriptor columns (BT2_F layout, cols 0..11 as BT_F):
+//  11 sp       padded polyfit slot count (s_pad(k))
+//  12 coeffoff byte offset of the f64 coeff block in the wire
+//  13 mapoff   byte offset of the mapping block in the wire
+//  14 nbits    mapping bit width (= bitlength(k-1))
+//  15 kmax     row stride of the padded sort matrix (globally uniform)
+// ===========================================================================
+
+#define PF_SMAX 23  // starts row stride (<= 22 slots + 1)
+#define bt2_row bt_row
+
+// scatter the bloom-ordered values into the padded [T, kmax] sort matrix
+// (pad = -inf so descending sort pushes pads to the tail)
+__global__ void bt2_padmat_kernel(const float* __restrict__ vals /*[K]*/,
+                                  const int64_t* __restrict__ desc, int nT,
+                                  int64_t K, int64_t kmax,
+                                  float* __restrict__ padmat) {
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < K; j += stride) {
+        int lo = 0, hi = nT - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt2_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
+        }
+        padmat[(int64_t)lo * kmax + (j - bt2_row(desc, lo)[3])] = vals[j];
+    }
+}
+
+__global__ void bt2_fill_ninf(float* __restrict__ p, int64_t n) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) p[i] = -INFINITY;
+}
+
+// per-tensor: derive padded segment starts from num_pos (same math as
+// polyfit_starts_kernel), write them to starts[T, PF_SMAX], and write the
+// num_pos trailer + the 5-byte mapping header into the wire
+__global__ void bt2_starts_kernel(const double* __restrict__ num_pos /*[T]*/,
+                                  const int64_t* __restrict__ desc,
+                                  int64_t* __restrict__ starts /*[T,PF_SMAX]*/,
+                                  uint8_t* __restrict__ wire, int d1) {
+    const int t = blockIdx.x;
+    if (threadIdx.x != 0) return;
+    const int64_t* D = bt2_row(desc, t);
+    const int64_t N = D[2];  // polyfit N = k (values per tensor)
+    const double RA[10] = {1.0 / 5, 1.0 / 10, 1.0 / 30, 1.0 / 100, 1.0 / 300,
+                           1.0 / 1000, 1.0 / 3000, 1.0 / 10000, 1.0 / 30000,
+                           1.0 / 100000};
+    const int NR = 10;
+    int NA = 0;
+    for (int i = 0; i < NR; ++i)
+        if ((int64_t)((double)N * RA[i]) > 30) NA = i + 1;
+    int64_t np = (int64_t)(num_pos[t] + 0.5);
+    int64_t nn = N - np;
+    int64_t pos[10], neg[10], psum = 0, nsum = 0;
+    for (int i = 0; i < NA; ++i) {
+        int64_t p = (int64_t)((double)np * RA[i]);
+        int64_t n = (int64_t)((double)nn * RA[i]);
+        pos[i] = (p > 30) ? p : 0;
+        neg[i] = (n > 30) ? n : 0;
+        psum += pos[i];
+        nsum += neg[i];
+    }
+    int64_t seg[2 * 10 + 2];
+    for (int i = 0; i < NA; ++i) seg[i] = pos[NA - 1 - i];
+    seg[NA] = np - psum;
+    seg[NA + 1] = nn - nsum;
+    for (int i = 0; i < NA; ++i) seg[NA + 2 + i] = neg[i];
+    int64_t* st = starts + (int64_t)t * PF_SMAX;
+    int64_t acc = 0;
+    st[0] = 0;
+    for (int i = 0; i < 2 * NA + 2; ++i) {
+        acc += seg[i];
+        st[i + 1] = acc;
+    }
+    // wire trailer: num_pos after the sp*d1 coefficients
+    const int64_t sp = D[11];
+    double* cw = (double*)(wire + D[12]);
+    cw[sp * d1] = (double)np;
+    // mapping header: count LE32 + nbits
+    uint8_t* mh = wire + D[13];
+    const int64_t k = D[2];
+    mh[0] = (uint8_t)(k & 255);
+    mh[1] = (uint8_t)((k >> 8) & 255);
+    mh[2] = (uint8_t)((k >> 16) & 255);
+    mh[3] = (uint8_t)((k >> 24) & 255);
+    mh[4] = (uint8_t)D[14];
+}
+
+// fit all (tensor, segment) pairs: grid = segmap length; y rows come from
+// the sorted pad matrix; coeffs are written straight into the wire
+__global__ void bt2_fit_kernel(const float* __restrict__ sorted /*[T,kmax]*/,
+                               const int64_t* __restrict__ desc,
+                               const int64_t* __restrict__ starts /*[T,PF_SMAX]*/,
+                               const int* __restrict__ seg_t,
+                               const int* __restrict__ seg_i, int degree,
+                               int64_t kmax, uint8_t* __restrict__ wire) {
+    const int t = seg_t[blockIdx.x];
+    const int si = seg_i[blockIdx.x];
+    const int64_t* D = bt2_row(desc, t);
+    const int64_t* st = starts + (int64_t)t * PF_SMAX;
+    const int64_t start = st[si];
+    const int64_t end = st[si + 1];
+    const int64_t len = end - start;
+    const int d1 = degree + 1;
+    const int np = 2 * degree + 1;
+    const double inv_len = len > 0 ? 1.0 / (double)len : 1.0;
+    const float* __restrict__ y = sorted + (int64_t)t * kmax;
+
+    double ps[2 * PF_MAXD1 - 1];
+    double mo[PF_MAXD1];
+    for (int p = 0; p < np; ++p) ps[p] = 0.0;
+    for (int p = 0; p < d1; ++p) mo[p] = 0.0;
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+        const double x = (double)(i - start + 1) * inv_len;
+        const double yv = (double)y[i];
+        double xp = 1.0;
+        for (int p = 0; p < np; ++p) {
+            ps[p] += xp;
+            if (p < d1) mo[p] += xp * yv;
+            xp *= x;
+        }
+    }
+    const int lane = threadIdx.x % WAVE, wid = threadIdx.x / WAVE;
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        for (int p = 0; p < np; ++p) ps[p] += __shfl_down(ps[p], off, WAVE);
+        for (int p = 0; p < d1; ++p) mo[p] += __shfl_down(mo[p], off, WAVE);
+    }
+    __shared__ double sh[QBLOCK / WAVE][2 * PF_MAXD1 - 1 + PF_MAXD1];
+    if (lane == 0) {
+        for (int p = 0; p < np; ++p) sh[wid][p] = ps[p];
+        for (int p = 0; p < d1; ++p) sh[wid][np + p] = mo[p];
+    }
+    __syncthreads();
+    if (threadIdx.x != 0) return;
+    for (int w = 1; w < (int)(blockDim.x / WAVE); ++w) {
+        for (int p = 0; p < np; ++p) ps[p] += sh[w][p];
+        for (int p = 0; p < d1; ++p) mo[p] += sh[w][np + p];
+    }
+    double G[PF_MAXD1][PF_MAXD1];
+    double dmax = 0.0;
+    for (int i = 0; i < d1; ++i) {
+        double dg = fabs(ps[2 * i]);
+        if (dg > dmax) dmax = dg;
+    }
+    const double ridge = dmax * 1e-10 + 1e-30;
+    for (int i = 0; i < d1; ++i)
+        for (int j = 0; j < d1; ++j)
+            G[i][j] = ps[i + j] + (i == j ? ridge : 0.0);
+    double L[PF_MAXD1][PF_MAXD1];
+    for (int i = 0; i < d1; ++i) {
+        for (int j = 0; j <= i; ++j) {
+            double sum = G[i][j];
+            for (int p = 0; p < j; ++p) sum -= L[i][p] * L[j][p];
+            if (i == j) L[i][j] = sqrt(sum > 1e-300 ? sum : 1e-300);
+            else L[i][j] = sum / L[j][j];
+        }
+    }
+    double yv[PF_MAXD1];
+    for (int i = 0; i < d1; ++i) {
+        double sum = mo[i];
+        for (int p = 0; p < i; ++p) sum -= L[i][p] * yv[p];
+        yv[i] = sum / L[i][i];
+    }
+    double* c = (double*)(wire + D[12]) + (int64_t)si * d1;
+    double cr[PF_MAXD1];
+    for (int i = d1 - 1; i >= 0; --i) {
+        double sum = yv[i];
+        for (int p = i + 1; p < d1; ++p) sum -= L[p][i] * cr[p];
+        cr[i] = sum / L[i][i];
+    }
+    for (int i = 0; i < d1; ++i) c[i] = cr[i];
+}
+
+// pack each tensor's mapping (argsort of the padded sort) at nbits into the
+// wire; bit positions can straddle words -> two atomicOr
+__global__ void bt2_pack_map_kernel(const int64_t* __restrict__ argsortm /*[T,kmax]*/,
+                                    const int64_t* __restrict__ desc, int nT,
+                                    int64_t K, int64_t kmax,
+                                    uint8_t* __restrict__ wire) {
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < K; j += stride) {
+        int lo = 0, hi = nT - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt2_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
+        }
+        const int64_t* D = bt2_row(desc, lo);
+        const int64_t local = j - D[3];
+        const uint64_t v = (uint64_t)argsortm[(int64_t)lo * kmax + local];
+        const int nb = (int)D[14];
+        // payload starts after the 5-byte header; base bit offset within
+        // the word-aligned region starting at mapoff
+        const int64_t bit0 = 40 + local * nb;
+        uint32_t* w = (uint32_t*)(wire + D[13]);
+        const int64_t wi = bit0 >> 5;
+        const int sh = (int)(bit0 & 31);
+        atomicOr(&w[wi], (uint32_t)(v << sh));
+        if (sh + nb > 32) atomicOr(&w[wi + 1], (uint32_t)(v >> (32 - sh)));
+    }
+}
+
+// evaluate the fitted curves: vals_eval[j] = poly(t, local position)
+__global__ void bt2_eval_kernel(const uint8_t* __restrict__ wire_r,
+                                const int64_t* __restrict__ desc, int nT,
+                                int64_t K,
+                                const int64_t* __restrict__ starts /*[T,PF_SMAX]*/,
+                                int d1, float* __restrict__ out) {
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < K; j += stride) {
+        int lo = 0, hi = nT - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt2_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
+        }
+        const int64_t* D = bt2_row(desc, lo);
+        const int64_t local = j - D[3];
+        const int64_t* st = starts + (int64_t)lo * PF_SMAX;
+        const int64_t sp = D[11];
+        int s = 0;
+        for (int i = 1; i < (int)sp; ++i)
+            if (st[i] <= local) s = i;
+        const int64_t sstart = st[s];
+        const int64_t slen = st[s + 1] - sstart;
+        const double x = (double)(local - sstart + 1) / (double)(slen > 0 ? slen : 1);
+        const double* c = (const double*)(wire_r + D[12]) + (int64_t)s * d1;
+        double yv = c[d1 - 1];
+        for (int p = d1 - 2; p >= 0; --p) yv = yv * x + c[p];
+        out[j] = (float)yv;
+    }
+}
+
+// derive starts for a (possibly remote) wire from its transmitted num_pos
+__global__ void bt2_starts_from_wire_kernel(const uint8_t* __restrict__ wire_r,
+                                            const int64_t* __restrict__ desc,
+                                            int d1,
+                                            int64_t* __restrict__ starts) {
+    const int t = blockIdx.x;
+    if (threadIdx.x != 0) return;
+    const int64_t* D = bt2_row(desc, t);
+    const double np = ((const double*)(wire_r + D[12]))[D[11] * d1];
+    const int64_t N = D[2];
+    const double RA[10] = {1.0 / 5, 1.0 / 10, 1.0 / 30, 1.0 / 100, 1.0 / 300,
+                           1.0 / 1000, 1.0 / 3000, 1.0 / 10000, 1.0 / 30000,
+                           1.0 / 100000};
+    const int NR = 10;
+    int NA = 0;
+    for (int i = 0; i < NR; ++i)
+        if ((int64_t)((double)N * RA[i]) > 30) NA = i + 1;
+    int64_t npi = (int64_t)(np + 0.5);
+    int64_t nn = N - npi;
+    int64_t pos[10], neg[10], psum = 0, nsum = 0;
+    for (int i = 0; i < NA; ++i) {
+        int64_t p = (int64_t)((double)npi * RA[i]);
+        int64_t n = (int64_t)((double)nn * RA[i]);
+        pos[i] = (p > 30) ? p : 0;
+        neg[i] = (n > 30) ? n : 0;
+        psum += pos[i];
+        nsum += neg[i];
+    }
+    int64_t seg[2 * 10 + 2];
+    for (int i = 0; i < NA; ++i) seg[i] = pos[NA - 1 - i];
+    seg[NA] = npi - psum;
+    seg[NA + 1] = nn - nsum;
+    for (int i = 0; i < NA; ++i) seg[NA + 2 + i] = neg[i];
+    int64_t* st = starts + (int64_t)t * PF_SMAX;
+    int64_t acc = 0;
+    st[0] = 0;
+    for (int i = 0; i < 2 * NA + 2; ++i) {
+        acc += seg[i];
+        st[i + 1] = acc;
+    }
+}
+
+// dense[voff + positives[koff + map(j)]] (+)= vals_eval[koff + j]
+// map(j) read straight out of the packed wire bits (no unpack pass)
+__global__ void bt2_scatter_both_kernel(const float* __restrict__ vals_eval,
+                                        const int64_t* __restrict__ positives,
+                                        const uint8_t* __restrict__ wire_r,
+                                        const int64_t* __restrict__ desc, int nT,
+                                        int64_t K, int accumulate,
+                                        float* __restrict__ dense) {
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < K; j += stride) {
+        int lo = 0, hi = nT - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt2_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
+        }
+        const int64_t* D = bt2_row(desc, lo);
+        const int64_t local = j - D[3];
+        const int nb = (int)D[14];
+        const int64_t bit0 = 40 + local * nb;
+        const uint32_t* w = (const uint32_t*)(wire_r + D[13]);
+        const int64_t wi = bit0 >> 5;
+        const int sh = (int)(bit0 & 31);
+        uint64_t acc = (uint64_t)w[wi] >> sh;
+        if (sh + nb > 32) acc |= (uint64_t)w[wi + 1] << (32 - sh);
+        const int64_t map = (int64_t)(acc & ((nb == 64) ? ~0ull : ((1ull << nb) - 1)));
+        const int64_t idx = positives[D[3] + map];
+        float* d = dense + D[1] + idx;
+        if (accumulate) *d += vals_eval[j];
+        else *d = vals_eval[j];
+    }
+}
+
+
+// ---------------------------------------------------------------------------
+// batched 'both' drivers
+// ---------------------------------------------------------------------------
+
+// Compress the whole model in 'both' mode: (wire, own_dense).
+std::vector<torch::Tensor> batched_compress_both(
+        torch::Tensor values_flat, torch::Tensor desc, torch::Tensor b2t,
+        torch::Tensor seg_t, torch::Tensor seg_i, int64_t wire_bytes,
+        int64_t k_total, int64_t mask_words, int64_t kmax, int64_t degree,
+        int64_t total_values) {
+    CHECK_CUDA(values_flat);
+    auto v = values_flat.contiguous();
+    auto d = desc.contiguous();
+    auto map = b2t.contiguous();
+    auto st_t = seg_t.contiguous();
+    auto st_i = seg_i.contiguous();
+    const int T = (int)d.size(0);
+    const int64_t BV = map.numel();
+    const int SB = (int)st_t.numel();
+    const int d1 = (int)degree + 1;
+    auto dev = v.device();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+
+    auto ws = torch::empty({(int64_t)T * (2 * TK_BINS + 4) + 6 * BV},
+                           torch::dtype(torch::kInt32).device(dev));
+    int* hist1 = ws.data_ptr<int>();
+    int* hist2 = hist1 + (int64_t)T * TK_BINS;
+    int* sc = hist2 + (int64_t)T * TK_BINS;
+    int* counts = sc + (int64_t)T * 4;
+    int* offs = counts + 2 * BV;
+    int* qcounts = offs + 2 * BV;
+    int* qoffs = qcounts + BV;
+    auto mask = torch::empty({mask_words}, torch::dtype(torch::kInt64).device(dev));
+    auto wire = torch::empty({wire_bytes}, torch::dtype(torch::kUInt8).device(dev));
+    auto out_idx = torch::empty({k_total}, torch::dtype(torch::kInt64).device(dev));
+    auto vals_tmp = torch::empty({k_total}, torch::dtype(torch::kFloat32).device(dev));
+    auto padmat = torch::empty({T, kmax}, torch::dtype(torch::kFloat32).device(dev));
+    auto starts = torch::empty({(int64_t)T * PF_SMAX},
+                               torch::dtype(torch::kInt64).device(dev));
+
+    zero_ints(hist1, (int64_t)T * (2 * TK_BINS + 4), stream);
+    zero_ints((int*)wire.data_ptr<uint8_t>(), wire_bytes / 4, stream);
+
+    const int64_t* dp = d.data_ptr<int64_t>();
+    const int* mp = map.data_ptr<int>();
+    const float* vp = v.data_ptr<float>();
+
+    hipLaunchKernelGGL(bt_hist_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, 1, hist1, sc);
+    hipLaunchKernelGGL(bt_thresh_kernel, dim3(T), dim3(WAVE), 0, stream, hist1, dp, 1, sc);
+    hipLaunchKernelGGL(bt_hist_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, 2, hist2, sc);
+    hipLaunchKernelGGL(bt_thresh_kernel, dim3(T), dim3(WAVE), 0, stream, hist2, dp, 2, sc);
+    hipLaunchKernelGGL(bt_count_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, sc, BV, counts);
+    hipLaunchKernelGGL(bt_scan_kernel, dim3(2 * T), dim3(QBLOCK), 0, stream,
+                       counts, dp, T, BV, offs);
+    hipLaunchKernelGGL(bt_scatter_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, sc, offs, BV, out_idx.data_ptr<int64_t>());
+    hipLaunchKernelGGL(bt_insert_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                       out_idx.data_ptr<int64_t>(), dp, T, k_total,
+                       wire.data_ptr<uint8_t>());
+    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+                       wire.data_ptr<uint8_t>(), wire_bytes, 1, dp, mp, BV,
+                       mask_words, qcounts, (uint64_t*)mask.data_ptr<int64_t>());
+    hipLaunchKernelGGL(bt_scan_kernel, dim3(T), dim3(QBLOCK), 0, stream,
+                       qcounts, dp, T, BV, qoffs);
+    // FP-aware gather to the temp buffer; positives into out_idx
+    hipLaunchKernelGGL(bt_qscatter_own_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+                       (const uint64_t*)mask.data_ptr<int64_t>(), qoffs, vp, dp, mp,
+                       (uint8_t*)nullptr, vals_tmp.data_ptr<float>(),
+                       out_idx.data_ptr<int64_t>());
+    // padded descending sort (stable: deterministic tie order)
+    hipLaunchKernelGGL(bt2_fill_ninf, dim3(bt_grid((int64_t)T * kmax)), dim3(256), 0,
+                       stream, padmat.data_ptr<float>(), (int64_t)T * kmax);
+    hipLaunchKernelGGL(bt2_padmat_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                       vals_tmp.data_ptr<float>(), dp, T, k_total, kmax,
+                       padmat.data_ptr<float>());
+    auto sorted_arg = at::sort(padmat, /*stable=*/true, /*dim=*/1, /*descending=*/true);
+    auto sorted = std::get<0>(sorted_arg).contiguous();
+    auto argsortm = std::get<1>(sorted_arg).contiguous();
+    auto num_pos = (sorted > 0).sum(1).to(torch::kFloat64).contiguous();
+    hipLaunchKernelGGL(bt2_starts_kernel, dim3(T), dim3(WAVE), 0, stream,
+                       num_pos.data_ptr<double>(), dp,
+                       starts.data_ptr<int64_t>(), wire.data_ptr<uint8_t>(), d1);
+    hipLaunchKernelGGL(bt2_fit_kernel, dim3(SB), dim3(QBLOCK), 0, stream,
+                       sorted.data_ptr<float>(), dp, starts.data_ptr<int64_t>(),
+                       st_t.data_ptr<int>(), st_i.data_ptr<int>(), (int)degree,
+                       kmax, wire.data_ptr<uint8_t>());
+    hipLaunchKernelGGL(bt2_pack_map_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                       argsortm.data_ptr<int64_t>(), dp, T, k_total, kmax,
+                       wire.data_ptr<uint8_t>());
+    // own decode: eval the fitted curves and scatter through the mapping
+    auto vals_eval = torch::empty({k_total}, torch::dtype(torch::kFloat32).device(dev));
+    auto own = torch::empty({total_values}, torch::dtype(torch::kFloat32).device(dev));
+    hipLaunchKernelGGL(bt_fill_zero_f, dim3(bt_grid(total_values)), dim3(256), 0, stream,
+                       own.data_ptr<float>(), total_values);
+    hipLaunchKernelGGL(bt2_eval_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                       wire.data_ptr<uint8_t>(), dp, T, k_total,
+                       starts.data_ptr<int64_t>(), d1, vals_eval.data_ptr<float>());
+    hipLaunchKernelGGL(bt2_scatter_both_kernel, dim3(bt_grid(k_total)), dim3(256), 0,
+                       stream, vals_eval.data_ptr<float>(),
+                       out_idx.data_ptr<int64_t>(), wire.data_ptr<uint8_t>(), dp, T,
+                       k_total, 0, own.data_ptr<float>());
+    return {wire, own};
+}
+
+// Multi-rank 'both' decode: SUM of dense decodes of R stacked wires.
+torch::Tensor batched_decode_both_sum(torch::Tensor wires2d, torch::Tensor desc,
+                                      torch::Tensor b2t, int64_t total_values,
+                                      int64_t mask_words, int64_t k_total,
+                                      int64_t degree) {
+    CHECK_CUDA(wires2d);
+    TORCH_CHECK(wires2d.dim() == 2, "expected [R, W]");
+    auto w = wires2d.contiguous();
+    auto d = desc.contiguous();
+    auto map = b2t.contiguous();
+    const int R = (int)w.size(0);
+    TORCH_CHECK(R >= 1 && R <= MAXR, "1..16 ranks supported");
+    const int64_t W = w.size(1);
+    const int T = (int)d.size(0);
+    const int64_t BV = map.numel();
+    const int d1 = (int)degree + 1;
+    auto dev = w.device();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+
+    auto ws = torch::empty({2 * (int64_t)R * BV}, torch::dtype(torch::kInt32).device(dev));
+    int* qcounts = ws.data_ptr<int>();
+    int* qoffs = qcounts + (int64_t)R * BV;
+    auto mask = torch::empty({(int64_t)R * mask_words},
+                             torch::dtype(torch::kInt64).device(dev));
+    auto dense = torch::empty({total_values}, torch::dtype(torch::kFloat32).device(dev));
+    auto positives = torch::empty({k_total}, torch::dtype(torch::kInt64).device(dev));
+    auto vals_eval = torch::empty({k_total}, torch::dtype(torch::kFloat32).device(dev));
+    auto starts = torch::empty({(int64_t)T * PF_SMAX},
+                               torch::dtype(torch::kInt64).device(dev));
+    hipLaunchKernelGGL(bt_fill_zero_f, dim3(bt_grid(total_values)), dim3(256), 0, stream,
+                       dense.data_ptr<float>(), total_values);
+    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+                       w.data_ptr<uint8_t>(), W, R, d.data_ptr<int64_t>(),
+                       map.data_ptr<int>(), BV, mask_words, qcounts,
+                       (uint64_t*)mask.data_ptr<int64_t>());
+    hipLaunchKernelGGL(bt_scan_kernel, dim3(R * T), dim3(QBLOCK), 0, stream,
+                       qcounts, d.data_ptr<int64_t>(), T, BV, qoffs);
+    for (int r = 0; r < R; ++r) {
+        const uint8_t* wr = w.data_ptr<uint8_t>() + (int64_t)r * W;
+        hipLaunchKernelGGL(bt_qscatter_own_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+                           (const uint64_t*)mask.data_ptr<int64_t>() + (int64_t)r * mask_words,
+                           qoffs + (int64_t)r * BV, (const float*)nullptr,
+                           d.data_ptr<int64_t>(), map.data_ptr<int>(),
+                           (uint8_t*)nullptr, (float*)nullptr,
+                           positives.data_ptr<int64_t>());
+        hipLaunchKernelGGL(bt2_starts_from_wire_kernel, dim3(T), dim3(WAVE), 0, stream,
+                           wr, d.data_ptr<int64_t>(), d1, starts.data_ptr<int64_t>());
+        hipLaunchKernelGGL(bt2_eval_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                           wr, d.data_ptr<int64_t>(), T, k_total,
+                           starts.data_ptr<int64_t>(), d1, vals_eval.data_ptr<float>());
+        hipLaunchKernelGGL(bt2_scatter_both_kernel, dim3(bt_grid(k_total)), dim3(256), 0,
+                           stream, vals_eval.data_ptr<float>(),
+                           positives.data_ptr<int64_t>(), wr, d.data_ptr<int64_t>(), T,
+                           k_total, 1, dense.data_ptr<float>());
+    }
+    return dense;
+}
+
 // ---------------------------------------------------------------------------
 // CPU-native C++ paths (replace the reference's TF C++ CPU ops:
 // bloom_filter_compression.cc / integer_compression.cc) — same wire format
@@ -1806,6 +2293,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("polyfit_eval", &polyfit_eval, "fused piecewise Horner eval");
     m.def("polyfit_starts", &polyfit_starts,
           "device-side padded segment boundaries from num_pos");
+    m.def("batched_compress_both", &batched_compress_both,
+          "whole-model 'both' compress: bloom+polyfit+packed mapping -> (wire, own_dense)");
+    m.def("batched_decode_both_sum", &batched_decode_both_sum,
+          "multi-rank 'both' decode: [R, W] wires -> sum of dense decodes");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
     m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");

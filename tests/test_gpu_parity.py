@@ -445,3 +445,43 @@ def test_graph_vs_eager_both_mode(dev):
     graphed = run(True)
     for a, b in zip(eager, graphed):
         assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
+
+
+def test_batched_both_pipeline_matches_per_tensor(dev):
+    """BothPipeline (whole-model bloom+polyfit+mapping) must match the
+    per-tensor DeepReduce wrapper path: same wire bytes, same training
+    results across steps (residual evolution included)."""
+    from deepreduce_amd import deepreduce_from_params
+
+    params = {
+        "compressor": "topk", "memory": "residual",
+        "communicator": "allgather", "compress_ratio": 0.01,
+        "deepreduce": "both", "index": "bloom", "policy": "leftmost",
+        "value": "polyfit",
+    }
+    grc_b = deepreduce_from_params(dict(params))
+    grc_p = deepreduce_from_params(dict(params))
+    torch.manual_seed(77)
+    named = [
+        ("w1", torch.randn(200_000, device=dev)),
+        ("w2", torch.randn(48_000, device=dev)),
+        ("w3", torch.randn(4_000, device=dev)),
+    ]
+    for step in range(3):
+        tensors = [(n, t * (1.0 + 0.3 * step)) for n, t in named]
+        fused = grc_b.step_many([(n, t.clone()) for n, t in tensors])
+        assert getattr(grc_b, "_bt_pipeline", None) is not None \
+            and grc_b._bt_pipeline[1].kind == "both", "BothPipeline did not engage"
+        total = 0
+        loop = []
+        for n, t in tensors:
+            loop.append(grc_p.step(t.clone(), n))
+            total += grc_p.last_wire_bytes
+        if step == 0:
+            assert grc_b.last_wire_bytes == total, \
+                (grc_b.last_wire_bytes, total)
+        for (n, _), f, l in zip(tensors, fused, loop):
+            d = (f.reshape(-1) - l.reshape(-1)).abs().max()
+            assert torch.allclose(f.reshape(-1), l.reshape(-1), atol=1e-5), \
+                f"step {step} tensor {n}: max diff {d}"
+    torch.cuda.synchronize()

@@ -260,3 +260,78 @@ def test_small_tensors_travel_dense_exact():
         p.join(timeout=60)
     for rank, ok in results:
         assert ok is True, f"rank {rank}: {ok}"
+
+
+def _run_overlap_parity(rank, world, q):
+    try:
+        _init(rank, world)
+        import torch.nn as nn
+
+        from deepreduce_amd import (DistributedOptimizer, broadcast_parameters,
+                                    deepreduce_from_params)
+        from deepreduce_amd.parallel import OverlappedReducer
+
+        params = {
+            "compressor": "topk", "memory": "residual",
+            "communicator": "allgather", "compress_ratio": 0.05,
+            "deepreduce": "index", "index": "bloom", "policy": "leftmost",
+        }
+
+        def make_model():
+            torch.manual_seed(1)
+            return nn.Sequential(nn.Linear(600, 40), nn.ReLU(), nn.Linear(40, 4))
+
+        def batch(step):
+            g = torch.Generator().manual_seed(1000 + 13 * step + rank)
+            return (torch.randn(8, 600, generator=g),
+                    torch.randint(0, 4, (8,), generator=g))
+
+        # path A: synchronous DistributedOptimizer
+        model_a = make_model()
+        broadcast_parameters(model_a)
+        grc_a = deepreduce_from_params(dict(params))
+        opt_a = DistributedOptimizer(torch.optim.SGD(model_a.parameters(), lr=0.1),
+                                     grc_a, model_a)
+        for s in range(4):
+            x, y = batch(s)
+            opt_a.zero_grad(set_to_none=False)
+            torch.nn.functional.cross_entropy(model_a(x), y).backward()
+            opt_a.step()
+
+        # path B: hook-driven OverlappedReducer
+        model_b = make_model()
+        broadcast_parameters(model_b)
+        grc_b = deepreduce_from_params(dict(params))
+        reducer = OverlappedReducer(model_b, grc_b)
+        sgd_b = torch.optim.SGD(model_b.parameters(), lr=0.1)
+        for s in range(4):
+            x, y = batch(s)
+            sgd_b.zero_grad(set_to_none=False)
+            reducer.zero_wire_counter()
+            torch.nn.functional.cross_entropy(model_b(x), y).backward()
+            reducer.finalize()
+            sgd_b.step()
+
+        ok = all(
+            torch.allclose(pa, pb, atol=1e-5)
+            for pa, pb in zip(model_a.parameters(), model_b.parameters())
+        )
+        q.put((rank, bool(ok)))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+def test_overlapped_reducer_matches_sync_optimizer():
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    os.environ["MASTER_PORT"] = "29753"
+    procs = [ctx.Process(target=_run_overlap_parity, args=(r, world, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, ok in results:
+        assert ok is True, f"rank {rank}: {ok}"

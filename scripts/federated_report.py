@@ -1,0 +1,119 @@
+#!/usr/bin/env python3
+"""Federated-round data-volume report — paper Table 2 equivalent.
+
+Runs real federated rounds (deepreduce_amd.federated, paper Algorithm 2:
+bidirectional compression of model deltas / client gradient sums with
+error feedback) on a small MLP and accounts the exact S2C and C2S bytes
+per round for the paper's configurations (Top-r vs DR variants).
+
+Usage: python scripts/federated_report.py [--rounds 2] [--clients 4] [--out docs/FEDERATED.md]
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from deepreduce_amd import deepreduce_from_params
+from deepreduce_amd.federated import FederatedClient, FederatedServer, run_federated_round
+
+CONFIGS = [
+    ("dense (baseline)", {"compressor": "none", "memory": "none"}),
+    ("Top-r(10%)", {}),
+    ("DR-BF-P0", {"deepreduce": "index", "index": "bloom", "policy": "p0"}),
+    ("DR-FitPoly", {"deepreduce": "value", "value": "polyfit"}),
+    ("DR-QSGD", {"deepreduce": "value", "value": "qsgd"}),
+    ("DR-QSGD+BF-P0", {"deepreduce": "both", "value": "qsgd",
+                       "index": "bloom", "policy": "p0"}),
+    ("DR-FitPoly+BF", {"deepreduce": "both", "value": "polyfit",
+                       "index": "bloom", "policy": "leftmost"}),
+]
+
+
+def make_model():
+    torch.manual_seed(0)
+    return torch.nn.Sequential(
+        torch.nn.Linear(128, 256), torch.nn.ReLU(),
+        torch.nn.Linear(256, 128), torch.nn.ReLU(),
+        torch.nn.Linear(128, 10),
+    )
+
+
+def run(rounds: int, clients: int):
+    dense_bytes = sum(p.numel() * 4 for p in make_model().parameters())
+    rows = []
+    for label, extra in CONFIGS:
+        params = {"compressor": "topk", "memory": "residual",
+                  "communicator": "allgather", "compress_ratio": 0.10}
+        params.update(extra)
+        grc = deepreduce_from_params(params)
+        # warm the model so x_t != x_0 (server sends a real delta)
+        model = make_model()
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        g = torch.Generator().manual_seed(7)
+        for _ in range(2):
+            x = torch.randn(32, 128, generator=g)
+            y = torch.randint(0, 10, (32,), generator=g)
+            opt.zero_grad()
+            torch.nn.functional.cross_entropy(model(x), y).backward()
+            opt.step()
+
+        server = FederatedServer(model, grc.compressor, lr=0.5)
+        cls = [FederatedClient(model, grc.compressor) for _ in range(clients)]
+
+        def data_iter(seed):
+            def it():
+                gg = torch.Generator().manual_seed(seed)
+                for _ in range(2):
+                    yield (torch.randn(16, 128, generator=gg),
+                           torch.randint(0, 10, (16,), generator=gg))
+            return it
+
+        s2c = c2s = 0
+        for r in range(rounds):
+            a, b = run_federated_round(
+                server, cls, [data_iter(100 * r + i) for i in range(clients)])
+            s2c += a
+            c2s += b
+        s2c /= rounds
+        c2s /= rounds / 1.0
+        rows.append((label, int(s2c), int(c2s / clients),
+                     s2c / dense_bytes, c2s / clients / dense_bytes))
+    return dense_bytes, rows
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rounds", type=int, default=2)
+    ap.add_argument("--clients", type=int, default=4)
+    ap.add_argument("--out", default=None)
+    args = ap.parse_args()
+    dense_bytes, rows = run(args.rounds, args.clients)
+    lines = [
+        "# Federated-round data volume (paper Algorithm 2 / Table 2 equivalent)",
+        "",
+        f"MLP model, {args.clients} clients, Top-r 10% + bidirectional error "
+        f"feedback; dense model = {dense_bytes:,} bytes.  S2C = server "
+        "broadcast of the compressed model delta; C2S = one client's "
+        "compressed gradient-sum push.  (Paper Table 2 on an RNN: Top-r "
+        "0.2033 rel. volume vs DR-QSGD-BF-P0 0.0621.)",
+        "",
+        "| Config | S2C bytes | C2S bytes/client | S2C rel. | C2S rel. |",
+        "|---|---:|---:|---:|---:|",
+    ]
+    for label, s2c, c2s, rs, rc in rows:
+        lines.append(f"| {label} | {s2c:,} | {c2s:,} | {rs:.4f} | {rc:.4f} |")
+    text = "\n".join(lines) + "\n"
+    print(text)
+    if args.out:
+        os.makedirs(os.path.dirname(args.out), exist_ok=True)
+        with open(args.out, "w") as f:
+            f.write(text)
+
+
+if __name__ == "__main__":
+    main()

@@ -12,7 +12,6 @@ Shares the batched on-device fit/eval machinery with PolyFit.
 """
 from __future__ import annotations
 
-import torch
 
 from . import SparseCompressor
 from .polyfit import _eval_segments, _fit_segments

@@ -14,8 +14,6 @@ per-tensor path in communicator.step_many.
 """
 from __future__ import annotations
 
-import math
-
 import torch
 
 BT_CHUNK = 8192

@@ -30,7 +30,7 @@ def get_args():
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--batch", type=int, default=128, help="per-GPU batch size")
     p.add_argument("--model", default="resnet50",
-                   choices=["resnet50", "resnet20", "ncf", "bert"])
+                   choices=["resnet50", "resnet20", "ncf", "bert", "mobilenet", "rnn"])
     p.add_argument("--compress-ratio", type=float, default=0.01)
     p.add_argument("--deepreduce", default="index", choices=["none", "value", "index", "both", "dense"])
     p.add_argument("--value", default="polyfit")
@@ -63,12 +63,16 @@ def build_grc(args):
 
 
 def make_batch(args, device):
-    if args.model in ("resnet50", "resnet20"):
+    if args.model in ("resnet50", "resnet20", "mobilenet"):
         res = 224 if args.model == "resnet50" else 32
         ncls = 1000 if args.model == "resnet50" else 10
         x = torch.randn(args.batch, 3, res, res, device=device)
         y = torch.randint(0, ncls, (args.batch,), device=device)
         return (x,), y
+    if args.model == "rnn":
+        ids = torch.randint(0, 10_004, (args.batch, 20), device=device)
+        y = torch.randint(0, 10_004, (args.batch, 20), device=device)
+        return (ids,), y
     if args.model == "ncf":
         u = torch.randint(0, 138_493, (args.batch,), device=device)
         i = torch.randint(0, 26_744, (args.batch,), device=device)
@@ -102,7 +106,7 @@ def main():
 
     torch.manual_seed(1234 + rank)
     model = registry[args.model]().to(device)
-    if args.model in ("resnet50", "resnet20") and use_cuda:
+    if args.model in ("resnet50", "resnet20", "mobilenet") and use_cuda:
         model = model.to(memory_format=torch.channels_last)
     broadcast_parameters(model)
 
@@ -135,7 +139,7 @@ def main():
     )
 
     inputs, target = make_batch(args, device)
-    if args.model in ("resnet50", "resnet20") and use_cuda:
+    if args.model in ("resnet50", "resnet20", "mobilenet") and use_cuda:
         inputs = (inputs[0].to(memory_format=torch.channels_last),)
 
     amp_dtype = torch.bfloat16
@@ -145,7 +149,7 @@ def main():
         opt.zero_grad(set_to_none=False)
         with autocast:
             out = model(*inputs)
-            if args.model == "bert":
+            if args.model in ("bert", "rnn"):
                 loss = loss_fn(out.float().flatten(0, 1), target.flatten())
             else:
                 loss = loss_fn(out.float(), target.float() if args.model == "ncf" else target)
@@ -178,15 +182,16 @@ def main():
     elapsed = float(t.item())
 
     ms_per_step = elapsed / args.steps * 1000.0
-    if args.model in ("resnet50", "resnet20"):
+    if args.model in ("resnet50", "resnet20", "mobilenet"):
         unit, metric = "images/sec", "images/sec"
         value = args.batch * world * args.steps / elapsed
     elif args.model == "ncf":
         unit, metric = "samples/sec", "samples/sec"
         value = args.batch * world * args.steps / elapsed
     else:
+        seq = 128 if args.model == "bert" else 20
         unit, metric = "tokens/sec", "tokens/sec"
-        value = args.batch * 128 * world * args.steps / elapsed
+        value = args.batch * seq * world * args.steps / elapsed
 
     n_params = sum(p.numel() for p in model.parameters())
     dense_bytes = n_params * 4
@@ -209,7 +214,7 @@ def main():
                     "config": {
                         "model": args.model,
                         "global_batch": args.batch * world,
-                        "seq_len": 128 if args.model == "bert" else None,
+                        "seq_len": {"bert": 128, "rnn": 20}.get(args.model),
                         "parallelism": f"dp{world}",
                         "deepreduce": args.deepreduce,
                         "compress_ratio": args.compress_ratio,

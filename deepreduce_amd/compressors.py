@@ -75,7 +75,9 @@ class TopKCompressor(Compressor):
         from .ops import topk_select
 
         vals, idxs = topk_select(flat, k)
-        return (vals, idxs), shape
+        # int32 on the wire (universe < 2^31 always): 8 bytes/entry like
+        # the reference's int32 keys, not 12
+        return (vals, idxs.int()), shape
 
     def decompress(self, tensors, ctx):
         vals, idxs = tensors
@@ -99,7 +101,7 @@ class ThresholdCompressor(Compressor):
         mask = flat.abs() >= self.threshold
         idxs = mask.nonzero(as_tuple=False).reshape(-1)
         vals = flat[idxs]
-        return (vals, idxs), shape
+        return (vals, idxs.int()), shape
 
     decompress = TopKCompressor.decompress
 
@@ -129,7 +131,7 @@ class RandomKCompressor(Compressor):
         idxs = torch.randperm(numel, generator=g)[:k].to(flat.device)
         self.step += 1
         vals = flat[idxs]
-        return (vals, idxs), shape
+        return (vals, idxs.int()), shape
 
     decompress = TopKCompressor.decompress
 

@@ -34,8 +34,13 @@ CONFIGS = [
 ]
 
 
-def make_model():
+def make_model(kind="mlp"):
     torch.manual_seed(0)
+    if kind == "rnn":
+        # paper Table 2 config: LSTM next-word model, 10k vocab
+        from deepreduce_amd.models import RnnLM
+
+        return RnnLM()
     return torch.nn.Sequential(
         torch.nn.Linear(128, 256), torch.nn.ReLU(),
         torch.nn.Linear(256, 128), torch.nn.ReLU(),
@@ -43,8 +48,8 @@ def make_model():
     )
 
 
-def run(rounds: int, clients: int):
-    dense_bytes = sum(p.numel() * 4 for p in make_model().parameters())
+def run(rounds: int, clients: int, model_kind: str = "mlp"):
+    dense_bytes = sum(p.numel() * 4 for p in make_model(model_kind).parameters())
     rows = []
     for label, extra in CONFIGS:
         params = {"compressor": "topk", "memory": "residual",
@@ -52,14 +57,21 @@ def run(rounds: int, clients: int):
         params.update(extra)
         grc = deepreduce_from_params(params)
         # warm the model so x_t != x_0 (server sends a real delta)
-        model = make_model()
+        model = make_model(model_kind)
         opt = torch.optim.SGD(model.parameters(), lr=0.1)
         g = torch.Generator().manual_seed(7)
         for _ in range(2):
-            x = torch.randn(32, 128, generator=g)
-            y = torch.randint(0, 10, (32,), generator=g)
-            opt.zero_grad()
-            torch.nn.functional.cross_entropy(model(x), y).backward()
+            if model_kind == "rnn":
+                ids = torch.randint(0, 10_004, (8, 20), generator=g)
+                tgt = torch.randint(0, 10_004, (8, 20), generator=g)
+                opt.zero_grad()
+                torch.nn.functional.cross_entropy(
+                    model(ids).flatten(0, 1), tgt.flatten()).backward()
+            else:
+                x = torch.randn(32, 128, generator=g)
+                y = torch.randint(0, 10, (32,), generator=g)
+                opt.zero_grad()
+                torch.nn.functional.cross_entropy(model(x), y).backward()
             opt.step()
 
         server = FederatedServer(model, grc.compressor, lr=0.5)
@@ -69,14 +81,24 @@ def run(rounds: int, clients: int):
             def it():
                 gg = torch.Generator().manual_seed(seed)
                 for _ in range(2):
-                    yield (torch.randn(16, 128, generator=gg),
-                           torch.randint(0, 10, (16,), generator=gg))
+                    if model_kind == "rnn":
+                        yield (torch.randint(0, 10_004, (8, 20), generator=gg),
+                               torch.randint(0, 10_004, (8, 20), generator=gg))
+                    else:
+                        yield (torch.randn(16, 128, generator=gg),
+                               torch.randint(0, 10, (16,), generator=gg))
             return it
 
+        loss_fn = None
+        if model_kind == "rnn":
+            def loss_fn(out, tgt):
+                return torch.nn.functional.cross_entropy(
+                    out.flatten(0, 1), tgt.flatten())
         s2c = c2s = 0
         for r in range(rounds):
             a, b = run_federated_round(
-                server, cls, [data_iter(100 * r + i) for i in range(clients)])
+                server, cls, [data_iter(100 * r + i) for i in range(clients)],
+                loss_fn=loss_fn)
             s2c += a
             c2s += b
         s2c /= rounds
@@ -91,12 +113,13 @@ def main():
     ap.add_argument("--rounds", type=int, default=2)
     ap.add_argument("--clients", type=int, default=4)
     ap.add_argument("--out", default=None)
+    ap.add_argument("--model", default="mlp", choices=["mlp", "rnn"])
     args = ap.parse_args()
-    dense_bytes, rows = run(args.rounds, args.clients)
+    dense_bytes, rows = run(args.rounds, args.clients, args.model)
     lines = [
         "# Federated-round data volume (paper Algorithm 2 / Table 2 equivalent)",
         "",
-        f"MLP model, {args.clients} clients, Top-r 10% + bidirectional error "
+        f"{args.model.upper()} model, {args.clients} clients, Top-r 10% + bidirectional error "
         f"feedback; dense model = {dense_bytes:,} bytes.  S2C = server "
         "broadcast of the compressed model delta; C2S = one client's "
         "compressed gradient-sum push.  (Paper Table 2 on an RNN: Top-r "

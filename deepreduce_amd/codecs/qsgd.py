@@ -40,6 +40,14 @@ class QSGD(SparseCompressor):
         n = levels.numel()
         nbytes = _norm_bytes(norms)  # [nb, 4]
 
+        bits = int(quantum_num * 2 + 1).bit_length()
+        if params.get("qsgd_pack", False) and bits < 8:
+            # sub-byte level packing (paper Table 2 runs QSGD at 7 bits):
+            # wire = [levels+q packed at `bits` bits][norm bytes]
+            packed = ops.pack_ints((levels.to(torch.int64) + quantum_num), bits)
+            return torch.cat([packed.view(torch.int8),
+                              nbytes.reshape(-1)]), idxs, shape
+
         full = (nb - 1) if n % bucket_size else nb
         pieces = []
         if full > 0:
@@ -59,6 +67,18 @@ class QSGD(SparseCompressor):
         quantum_num = int(params.get("quantum_num", 127))
         bucket_size = int(params.get("bucket_size", 512))
         stride = bucket_size + 4
+
+        bits = int(quantum_num * 2 + 1).bit_length()
+        if params.get("qsgd_pack", False) and bits < 8:
+            n = int(idxs.numel())
+            nb = (n + bucket_size - 1) // bucket_size
+            nbits_bytes = (n * bits + 7) // 8
+            packed, nb8 = wire.split([nbits_bytes, 4 * nb])
+            levels = (ops.unpack_ints(packed.view(torch.uint8).contiguous(), n, bits)
+                      - quantum_num).to(torch.int8)
+            norms = _bytes_norm(nb8)
+            vals = ops.qsgd_dequantize(levels, norms, quantum_num, bucket_size)
+            return vals, idxs, shape
 
         total = wire.numel()
         full = total // stride

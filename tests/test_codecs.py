@@ -168,3 +168,28 @@ class TestPolySeg:
         assert fixed_segments(500, 10) == fixed_segments(500, 10)
         assert sum(fixed_segments(12345, 10)) == 12345
         assert sum(fixed_segments(7, 10)) == 7
+
+
+def test_qsgd_packed_levels_roundtrip():
+    """qsgd_pack: sub-byte level packing (7 bits at quantum 63) halves...
+    reduces the level bytes while reconstruction stays within the QSGD
+    error bound."""
+    import torch
+
+    from deepreduce_amd.codecs import compressor
+
+    torch.manual_seed(0)
+    vals = torch.randn(5000)
+    idxs = torch.arange(5000)
+    shape = torch.Size([100_000])
+    params = {"quantum_num": 63, "bucket_size": 512, "qsgd_pack": True}
+    w, i, s = compressor["qsgd"].compress((vals.clone(), idxs, shape), params)
+    params_u = dict(params, qsgd_pack=False)
+    w8, _, _ = compressor["qsgd"].compress((vals.clone(), idxs, shape), params_u)
+    assert w.numel() < w8.numel()  # packed is smaller
+    v2, i2, _ = compressor["qsgd"].decompress((w, i, s), params)
+    assert torch.equal(i2, idxs)
+    pad = (-5000) % 512
+    norms = torch.nn.functional.pad(vals, (0, pad)).view(-1, 512).norm(dim=1)
+    bound = (norms / 63 * 1.001 + 1e-6).repeat_interleave(512)[:5000]
+    assert ((v2 - vals).abs() <= bound).all()

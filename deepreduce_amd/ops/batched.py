@@ -189,6 +189,84 @@ class BothPipeline(BatchedPipeline):
         )
 
 
+class ValuePipeline(BatchedPipeline):
+    """Value-mode whole-model pipeline: polyfit coefficients + raw int32
+    indices in value-sorted order (no bloom), wire-compatible with the
+    generic ValueCompressor payload."""
+
+    kind = "value"
+
+    def __init__(self, names, numels, params, device):
+        from ..codecs.polyfit import s_pad
+
+        self.names = list(names)
+        self.numels = list(numels)
+        self.degree = int(params.get("poly_degree", 5))
+        d1 = self.degree + 1
+        T = len(numels)
+        ratio = params.get("compress_ratio", 0.01)
+
+        desc = torch.zeros(T, 16, dtype=torch.int64)
+        b2t, seg_t, seg_i = [], [], []
+        voff = koff = wire_off = cntoff = mwoff = blkoff = 0
+        kmax = 1
+        self.metas = []
+        for t, n in enumerate(numels):
+            k = max(1, int(round(n * ratio)))
+            kmax = max(kmax, k)
+            nb = (n + BT_CHUNK - 1) // BT_CHUNK
+            sp = s_pad(k)
+            coeff_bytes = (sp * d1 + 1) * 8
+            desc[t, 0] = n
+            desc[t, 1] = voff
+            desc[t, 2] = k
+            desc[t, 3] = koff
+            desc[t, 6] = wire_off + coeff_bytes   # int32 idxs
+            desc[t, 8] = cntoff
+            desc[t, 9] = mwoff
+            desc[t, 10] = blkoff
+            desc[t, 11] = sp
+            desc[t, 12] = wire_off                # coeffs
+            desc[t, 13] = -1                      # no mapping chunk
+            self.metas.append([(torch.float64, sp * d1 + 1), (torch.int32, k)])
+            b2t.extend([t] * nb)
+            seg_t.extend([t] * sp)
+            seg_i.extend(range(sp))
+            voff += n
+            koff += k
+            wire_off += coeff_bytes + _pad8(4 * k)
+            cntoff += nb
+            mwoff += nb * (BT_CHUNK // 64)
+            blkoff += nb
+
+        self.total_values = voff
+        self.k_total = koff
+        self.kmax = kmax
+        self.wire_bytes = wire_off
+        self.mask_words = mwoff
+        self.desc = desc.to(device)
+        self.b2t = torch.tensor(b2t, dtype=torch.int32, device=device)
+        self.seg_t = torch.tensor(seg_t, dtype=torch.int32, device=device)
+        self.seg_i = torch.tensor(seg_i, dtype=torch.int32, device=device)
+
+    def compress_and_own(self, values_flat: torch.Tensor):
+        from deepreduce_amd import _hip_ops
+
+        wire, own = _hip_ops.batched_compress_value(
+            values_flat, self.desc, self.b2t, self.seg_t, self.seg_i,
+            self.wire_bytes, self.k_total, self.kmax, self.degree,
+            self.total_values,
+        )
+        return wire, own
+
+    def decode_sum(self, wires2d):
+        from deepreduce_amd import _hip_ops
+
+        return _hip_ops.batched_decode_value_sum(
+            wires2d, self.desc, self.total_values, self.k_total, self.degree,
+        )
+
+
 def maybe_pipeline(communicator, comp, named_tensors):
     """Return a cached pipeline when the configuration qualifies:
     BatchedPipeline for IndexCompressor(topk+bloom+leftmost),
@@ -197,7 +275,7 @@ def maybe_pipeline(communicator, comp, named_tensors):
     from ..codecs.bloom import Bloom
     from ..codecs.polyfit import PolyFit
     from ..compressors import TopKCompressor
-    from ..wrappers import DeepReduce, IndexCompressor
+    from ..wrappers import DeepReduce, IndexCompressor, ValueCompressor
 
     cls = None
     if isinstance(comp, IndexCompressor) and comp.idx_codec is Bloom:
@@ -205,15 +283,20 @@ def maybe_pipeline(communicator, comp, named_tensors):
     elif (isinstance(comp, DeepReduce) and comp.idx_codec is Bloom
           and comp.val_codec is PolyFit):
         cls = BothPipeline
+    elif isinstance(comp, ValueCompressor) and comp.val_codec is PolyFit:
+        cls = ValuePipeline
     if cls is None:
         return None
     params = comp.params
-    if params.get("policy", "leftmost") != "leftmost" or params.get("micro-benchmark"):
+    if params.get("micro-benchmark"):
+        return None
+    if cls is not ValuePipeline and params.get("policy", "leftmost") != "leftmost":
+        return None
+    if cls in (BothPipeline, ValuePipeline) and params.get("sort", False):
         return None
     if cls is BothPipeline and (
         not params.get("pack_mapping", True)
         or not params.get("fp_aware", True)
-        or params.get("sort", False)
     ):
         return None
     if not isinstance(comp.sparsifier, TopKCompressor):

@@ -1062,7 +1062,8 @@ __global__ void bt_scatter_kernel(const float* __restrict__ vals,
                                   const int* __restrict__ b2t,
                                   const int* __restrict__ sc,
                                   const int* __restrict__ offs, int64_t BV,
-                                  int64_t* __restrict__ out_i) {
+                                  int64_t* __restrict__ out_i,
+                                  float* __restrict__ out_v /*nullable*/) {
     const int t = b2t[blockIdx.x];
     const int64_t* D = bt_row(desc, t);
     const int64_t lb = blockIdx.x - D[10];
@@ -1074,6 +1075,7 @@ __global__ void bt_scatter_kernel(const float* __restrict__ vals,
     const int64_t k = D[2];
     const int64_t need = k - above;
     int64_t* __restrict__ out = out_i + D[3];
+    float* __restrict__ ov = out_v ? out_v + D[3] : nullptr;
     int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
     __shared__ int wc0[TK_BLOCK / WAVE], wc1[TK_BLOCK / WAVE];
     __shared__ int base0_s, base1_s;
@@ -1095,12 +1097,17 @@ __global__ void bt_scatter_kernel(const float* __restrict__ vals,
         if (p0) {
             int wbase = 0;
             for (int w = 0; w < wid; ++w) wbase += wc0[w];
-            out[base0_s + wbase + __popcll(b0 & below)] = i;
+            int64_t oi = base0_s + wbase + __popcll(b0 & below);
+            out[oi] = i;
+            if (ov) ov[oi] = v[i];
         } else if (p1) {
             int wbase = 0;
             for (int w = 0; w < wid; ++w) wbase += wc1[w];
             int64_t ordinal = base1_s + wbase + __popcll(b1 & below);
-            if (ordinal < need) out[above + ordinal] = i;
+            if (ordinal < need) {
+                out[above + ordinal] = i;
+                if (ov) ov[above + ordinal] = v[i];
+            }
         }
         __syncthreads();
         if (threadIdx.x == 0) {
@@ -1384,7 +1391,8 @@ std::vector<torch::Tensor> batched_compress(torch::Tensor values_flat,
     hipLaunchKernelGGL(bt_scan_kernel, dim3(2 * T), dim3(QBLOCK), 0, stream,
                        counts, dp, T, BV, offs);
     hipLaunchKernelGGL(bt_scatter_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
-                       vp, dp, mp, sc, offs, BV, out_idx.data_ptr<int64_t>());
+                       vp, dp, mp, sc, offs, BV, out_idx.data_ptr<int64_t>(),
+                       (float*)nullptr);
     hipLaunchKernelGGL(bt_insert_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
                        out_idx.data_ptr<int64_t>(), dp, T, k_total,
                        wire.data_ptr<uint8_t>());
@@ -1763,14 +1771,17 @@ __global__ void bt2_starts_kernel(const double* __restrict__ num_pos /*[T]*/,
     const int64_t sp = D[11];
     double* cw = (double*)(wire + D[12]);
     cw[sp * d1] = (double)np;
-    // mapping header: count LE32 + nbits
-    uint8_t* mh = wire + D[13];
-    const int64_t k = D[2];
-    mh[0] = (uint8_t)(k & 255);
-    mh[1] = (uint8_t)((k >> 8) & 255);
-    mh[2] = (uint8_t)((k >> 16) & 255);
-    mh[3] = (uint8_t)((k >> 24) & 255);
-    mh[4] = (uint8_t)D[14];
+    // mapping header: count LE32 + nbits (both-mode only; value-mode
+    // descriptors carry D[13] < 0)
+    if (D[13] >= 0) {
+        uint8_t* mh = wire + D[13];
+        const int64_t k = D[2];
+        mh[0] = (uint8_t)(k & 255);
+        mh[1] = (uint8_t)((k >> 8) & 255);
+        mh[2] = (uint8_t)((k >> 16) & 255);
+        mh[3] = (uint8_t)((k >> 24) & 255);
+        mh[4] = (uint8_t)D[14];
+    }
 }
 
 // fit all (tensor, segment) pairs: grid = segmap length; y rows come from
@@ -2052,7 +2063,8 @@ std::vector<torch::Tensor> batched_compress_both(
     hipLaunchKernelGGL(bt_scan_kernel, dim3(2 * T), dim3(QBLOCK), 0, stream,
                        counts, dp, T, BV, offs);
     hipLaunchKernelGGL(bt_scatter_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
-                       vp, dp, mp, sc, offs, BV, out_idx.data_ptr<int64_t>());
+                       vp, dp, mp, sc, offs, BV, out_idx.data_ptr<int64_t>(),
+                       (float*)nullptr);
     hipLaunchKernelGGL(bt_insert_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
                        out_idx.data_ptr<int64_t>(), dp, T, k_total,
                        wire.data_ptr<uint8_t>());
@@ -2155,6 +2167,176 @@ torch::Tensor batched_decode_both_sum(torch::Tensor wires2d, torch::Tensor desc,
                            stream, vals_eval.data_ptr<float>(),
                            positives.data_ptr<int64_t>(), wr, d.data_ptr<int64_t>(), T,
                            k_total, 1, dense.data_ptr<float>());
+    }
+    return dense;
+}
+
+
+
+// ---------------------------------------------------------------------------
+// batched 'value' pipeline (polyfit coefficients + raw int32 indices in
+// value-sorted order — no bloom).  Wire per tensor, matching the generic
+// ValueCompressor payload: [coeffs f64 (sp*d1+1), pad8][idxs int32 k, pad8].
+// Descriptor col 6 = idxoff (bytes); col 13 = -1 (no mapping header).
+// ---------------------------------------------------------------------------
+
+// wire_idx[ord] = out_idx[koff + argsort(ord)] as int32
+__global__ void bt2_write_idx_kernel(const int64_t* __restrict__ out_idx,
+                                     const int64_t* __restrict__ argsortm,
+                                     const int64_t* __restrict__ desc, int nT,
+                                     int64_t K, int64_t kmax,
+                                     uint8_t* __restrict__ wire) {
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < K; j += stride) {
+        int lo = 0, hi = nT - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
+        }
+        const int64_t* D = bt_row(desc, lo);
+        const int64_t local = j - D[3];
+        const int64_t src = argsortm[(int64_t)lo * kmax + local];
+        ((int32_t*)(wire + D[6]))[local] = (int32_t)out_idx[D[3] + src];
+    }
+}
+
+// dense[voff + wire_idx[ord]] (+)= vals_eval[ord]
+__global__ void bt2_scatter_value_kernel(const float* __restrict__ vals_eval,
+                                         const uint8_t* __restrict__ wire_r,
+                                         const int64_t* __restrict__ desc,
+                                         int nT, int64_t K, int accumulate,
+                                         float* __restrict__ dense) {
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < K; j += stride) {
+        int lo = 0, hi = nT - 1;
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
+        }
+        const int64_t* D = bt_row(desc, lo);
+        const int64_t local = j - D[3];
+        const int64_t idx = (int64_t)((const int32_t*)(wire_r + D[6]))[local];
+        float* d = dense + D[1] + idx;
+        if (accumulate) *d += vals_eval[j];
+        else *d = vals_eval[j];
+    }
+}
+
+std::vector<torch::Tensor> batched_compress_value(
+        torch::Tensor values_flat, torch::Tensor desc, torch::Tensor b2t,
+        torch::Tensor seg_t, torch::Tensor seg_i, int64_t wire_bytes,
+        int64_t k_total, int64_t kmax, int64_t degree, int64_t total_values) {
+    CHECK_CUDA(values_flat);
+    auto v = values_flat.contiguous();
+    auto d = desc.contiguous();
+    auto map = b2t.contiguous();
+    auto st_t = seg_t.contiguous();
+    auto st_i = seg_i.contiguous();
+    const int T = (int)d.size(0);
+    const int64_t BV = map.numel();
+    const int SB = (int)st_t.numel();
+    const int d1 = (int)degree + 1;
+    auto dev = v.device();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+
+    auto ws = torch::empty({(int64_t)T * (2 * TK_BINS + 4) + 4 * BV},
+                           torch::dtype(torch::kInt32).device(dev));
+    int* hist1 = ws.data_ptr<int>();
+    int* hist2 = hist1 + (int64_t)T * TK_BINS;
+    int* sc = hist2 + (int64_t)T * TK_BINS;
+    int* counts = sc + (int64_t)T * 4;
+    int* offs = counts + 2 * BV;
+    auto wire = torch::empty({wire_bytes}, torch::dtype(torch::kUInt8).device(dev));
+    auto out_idx = torch::empty({k_total}, torch::dtype(torch::kInt64).device(dev));
+    auto vals_tmp = torch::empty({k_total}, torch::dtype(torch::kFloat32).device(dev));
+    auto padmat = torch::empty({T, kmax}, torch::dtype(torch::kFloat32).device(dev));
+    auto starts = torch::empty({(int64_t)T * PF_SMAX},
+                               torch::dtype(torch::kInt64).device(dev));
+    zero_ints(hist1, (int64_t)T * (2 * TK_BINS + 4), stream);
+    zero_ints((int*)wire.data_ptr<uint8_t>(), wire_bytes / 4, stream);
+
+    const int64_t* dp = d.data_ptr<int64_t>();
+    const int* mp = map.data_ptr<int>();
+    const float* vp = v.data_ptr<float>();
+    hipLaunchKernelGGL(bt_hist_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, 1, hist1, sc);
+    hipLaunchKernelGGL(bt_thresh_kernel, dim3(T), dim3(WAVE), 0, stream, hist1, dp, 1, sc);
+    hipLaunchKernelGGL(bt_hist_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, 2, hist2, sc);
+    hipLaunchKernelGGL(bt_thresh_kernel, dim3(T), dim3(WAVE), 0, stream, hist2, dp, 2, sc);
+    hipLaunchKernelGGL(bt_count_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, sc, BV, counts);
+    hipLaunchKernelGGL(bt_scan_kernel, dim3(2 * T), dim3(QBLOCK), 0, stream,
+                       counts, dp, T, BV, offs);
+    hipLaunchKernelGGL(bt_scatter_kernel, dim3((int)BV), dim3(TK_BLOCK), 0, stream,
+                       vp, dp, mp, sc, offs, BV, out_idx.data_ptr<int64_t>(),
+                       vals_tmp.data_ptr<float>());
+    hipLaunchKernelGGL(bt2_fill_ninf, dim3(bt_grid((int64_t)T * kmax)), dim3(256), 0,
+                       stream, padmat.data_ptr<float>(), (int64_t)T * kmax);
+    hipLaunchKernelGGL(bt2_padmat_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                       vals_tmp.data_ptr<float>(), dp, T, k_total, kmax,
+                       padmat.data_ptr<float>());
+    auto sorted_arg = at::sort(padmat, /*stable=*/true, /*dim=*/1, /*descending=*/true);
+    auto sorted = std::get<0>(sorted_arg).contiguous();
+    auto argsortm = std::get<1>(sorted_arg).contiguous();
+    auto num_pos = (sorted > 0).sum(1).to(torch::kFloat64).contiguous();
+    hipLaunchKernelGGL(bt2_starts_kernel, dim3(T), dim3(WAVE), 0, stream,
+                       num_pos.data_ptr<double>(), dp,
+                       starts.data_ptr<int64_t>(), wire.data_ptr<uint8_t>(), d1);
+    hipLaunchKernelGGL(bt2_fit_kernel, dim3(SB), dim3(QBLOCK), 0, stream,
+                       sorted.data_ptr<float>(), dp, starts.data_ptr<int64_t>(),
+                       st_t.data_ptr<int>(), st_i.data_ptr<int>(), (int)degree,
+                       kmax, wire.data_ptr<uint8_t>());
+    hipLaunchKernelGGL(bt2_write_idx_kernel, dim3(bt_grid(k_total)), dim3(256), 0,
+                       stream, out_idx.data_ptr<int64_t>(),
+                       argsortm.data_ptr<int64_t>(), dp, T, k_total, kmax,
+                       wire.data_ptr<uint8_t>());
+    // own decode
+    auto vals_eval = torch::empty({k_total}, torch::dtype(torch::kFloat32).device(dev));
+    auto own = torch::empty({total_values}, torch::dtype(torch::kFloat32).device(dev));
+    hipLaunchKernelGGL(bt_fill_zero_f, dim3(bt_grid(total_values)), dim3(256), 0, stream,
+                       own.data_ptr<float>(), total_values);
+    hipLaunchKernelGGL(bt2_eval_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                       wire.data_ptr<uint8_t>(), dp, T, k_total,
+                       starts.data_ptr<int64_t>(), d1, vals_eval.data_ptr<float>());
+    hipLaunchKernelGGL(bt2_scatter_value_kernel, dim3(bt_grid(k_total)), dim3(256), 0,
+                       stream, vals_eval.data_ptr<float>(), wire.data_ptr<uint8_t>(),
+                       dp, T, k_total, 0, own.data_ptr<float>());
+    return {wire, own};
+}
+
+torch::Tensor batched_decode_value_sum(torch::Tensor wires2d, torch::Tensor desc,
+                                       int64_t total_values, int64_t k_total,
+                                       int64_t degree) {
+    CHECK_CUDA(wires2d);
+    TORCH_CHECK(wires2d.dim() == 2, "expected [R, W]");
+    auto w = wires2d.contiguous();
+    auto d = desc.contiguous();
+    const int R = (int)w.size(0);
+    const int64_t W = w.size(1);
+    const int T = (int)d.size(0);
+    const int d1 = (int)degree + 1;
+    auto dev = w.device();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    auto dense = torch::empty({total_values}, torch::dtype(torch::kFloat32).device(dev));
+    auto vals_eval = torch::empty({k_total}, torch::dtype(torch::kFloat32).device(dev));
+    auto starts = torch::empty({(int64_t)T * PF_SMAX},
+                               torch::dtype(torch::kInt64).device(dev));
+    hipLaunchKernelGGL(bt_fill_zero_f, dim3(bt_grid(total_values)), dim3(256), 0, stream,
+                       dense.data_ptr<float>(), total_values);
+    for (int r = 0; r < R; ++r) {
+        const uint8_t* wr = w.data_ptr<uint8_t>() + (int64_t)r * W;
+        hipLaunchKernelGGL(bt2_starts_from_wire_kernel, dim3(T), dim3(WAVE), 0, stream,
+                           wr, d.data_ptr<int64_t>(), d1, starts.data_ptr<int64_t>());
+        hipLaunchKernelGGL(bt2_eval_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
+                           wr, d.data_ptr<int64_t>(), T, k_total,
+                           starts.data_ptr<int64_t>(), d1, vals_eval.data_ptr<float>());
+        hipLaunchKernelGGL(bt2_scatter_value_kernel, dim3(bt_grid(k_total)), dim3(256), 0,
+                           stream, vals_eval.data_ptr<float>(), wr,
+                           d.data_ptr<int64_t>(), T, k_total, 1,
+                           dense.data_ptr<float>());
     }
     return dense;
 }
@@ -2297,6 +2479,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "whole-model 'both' compress: bloom+polyfit+packed mapping -> (wire, own_dense)");
     m.def("batched_decode_both_sum", &batched_decode_both_sum,
           "multi-rank 'both' decode: [R, W] wires -> sum of dense decodes");
+    m.def("batched_compress_value", &batched_compress_value,
+          "whole-model value-mode compress: polyfit coeffs + int32 idxs");
+    m.def("batched_decode_value_sum", &batched_decode_value_sum,
+          "multi-rank value-mode decode");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
     m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");

@@ -485,3 +485,37 @@ def test_batched_both_pipeline_matches_per_tensor(dev):
             assert torch.allclose(f.reshape(-1), l.reshape(-1), atol=1e-5), \
                 f"step {step} tensor {n}: max diff {d}"
     torch.cuda.synchronize()
+
+
+def test_batched_value_pipeline_matches_per_tensor(dev):
+    """ValuePipeline (polyfit coeffs + int32 idxs, no bloom) vs the
+    per-tensor ValueCompressor path: same wire bytes, same results."""
+    from deepreduce_amd import deepreduce_from_params
+
+    params = {
+        "compressor": "topk", "memory": "residual",
+        "communicator": "allgather", "compress_ratio": 0.01,
+        "deepreduce": "value", "value": "polyfit",
+    }
+    grc_b = deepreduce_from_params(dict(params))
+    grc_p = deepreduce_from_params(dict(params))
+    torch.manual_seed(88)
+    named = [("w1", torch.randn(150_000, device=dev)),
+             ("w2", torch.randn(30_000, device=dev)),
+             ("w3", torch.randn(5_500, device=dev))]
+    for step in range(3):
+        tensors = [(n, t * (1.0 + 0.2 * step)) for n, t in named]
+        fused = grc_b.step_many([(n, t.clone()) for n, t in tensors])
+        assert getattr(grc_b, "_bt_pipeline", None) is not None \
+            and grc_b._bt_pipeline[1].kind == "value", "ValuePipeline did not engage"
+        total = 0
+        loop = []
+        for n, t in tensors:
+            loop.append(grc_p.step(t.clone(), n))
+            total += grc_p.last_wire_bytes
+        if step == 0:
+            assert grc_b.last_wire_bytes == total, (grc_b.last_wire_bytes, total)
+        for (n, _), f, l in zip(tensors, fused, loop):
+            assert torch.allclose(f.reshape(-1), l.reshape(-1), atol=1e-5), \
+                f"step {step} {n}: {(f - l).abs().max()}"
+    torch.cuda.synchronize()

@@ -36,6 +36,7 @@ def get_args():
     p.add_argument("--value", default="polyfit")
     p.add_argument("--index", default="bloom")
     p.add_argument("--policy", default="leftmost")
+    p.add_argument("--wire-dtype", default="fp32", choices=["fp32", "fp16"])
     p.add_argument("--device", default=None)
     p.add_argument("--overlap", action="store_true",
                    help="hook-driven compression overlapped with backward + "
@@ -58,6 +59,7 @@ def build_grc(args):
             "value": args.value,
             "index": args.index,
             "policy": args.policy,
+            "wire_dtype": args.wire_dtype,
         }
     return deepreduce_from_params(params), params
 

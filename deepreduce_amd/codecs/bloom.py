@@ -133,9 +133,14 @@ class Bloom(SparseCompressor):
                 positives = ops.bloom_query_positives(packed, m, num_hash, grad_size)
                 new_idxs = _policy_select(positives, num_indices, policy, params, m, num_hash)
             vals = dense.reshape(-1)[new_idxs]
+            if params.get("wire_dtype") == "fp16":
+                # half-precision wire values: the residual is computed from
+                # the DECODED payload, so error feedback absorbs the
+                # quantization (same mechanism as QSGD)
+                vals = vals.half()
             # side-channel for the wrappers' own-payload cache: decompress of
             # this payload deterministically yields exactly (vals, new_idxs)
-            params["_own_decoded"] = (vals, new_idxs)
+            params["_own_decoded"] = (vals.float(), new_idxs)
 
         if policy == "p0":
             # True insert count travels IN-BAND at the tail of the bit
@@ -180,7 +185,7 @@ class Bloom(SparseCompressor):
                 vals = vals[:n]
             elif n > vals.numel():  # under-full positives (shouldn't happen)
                 idxs = idxs[: vals.numel()]
-        return vals, idxs, shape
+        return vals.float(), idxs, shape
 
 
 class BloomCPU(Bloom):

@@ -33,6 +33,9 @@ class BatchedPipeline:
 
         self.names = list(names)
         self.numels = list(numels)
+        self.wire_half = params.get("wire_dtype") == "fp16"
+        vb = 2 if self.wire_half else 4  # bytes per wire value
+        vdt = torch.float16 if self.wire_half else torch.float32
         T = len(numels)
         ratio = params.get("compress_ratio", 0.01)
 
@@ -51,16 +54,16 @@ class BatchedPipeline:
             desc[t, 3] = koff
             desc[t, 4] = m
             desc[t, 5] = num_hash
-            desc[t, 6] = wire_off + _pad8(4 * k)   # bits after padded vals
+            desc[t, 6] = wire_off + _pad8(vb * k)  # bits after padded vals
             desc[t, 7] = wire_off                  # vals first (payload order)
             desc[t, 8] = cntoff
             desc[t, 9] = mwoff
             desc[t, 10] = blkoff
-            self.metas.append([(torch.float32, k), (torch.uint8, nbytes)])
+            self.metas.append([(vdt, k), (torch.uint8, nbytes)])
             b2t.extend([t] * nb)
             voff += n
             koff += k
-            wire_off += _pad8(4 * k) + _pad8(nbytes)
+            wire_off += _pad8(vb * k) + _pad8(nbytes)
             cntoff += nb
             mwoff += nb * (BT_CHUNK // 64)
             blkoff += nb
@@ -78,7 +81,7 @@ class BatchedPipeline:
 
         wire, out_idx = _hip_ops.batched_compress(
             values_flat, self.desc, self.b2t, self.wire_bytes, self.k_total,
-            self.mask_words,
+            self.mask_words, int(self.wire_half),
         )
         return wire, out_idx
 
@@ -90,13 +93,15 @@ class BatchedPipeline:
         from deepreduce_amd import _hip_ops
 
         return _hip_ops.batched_scatter_dense(wire, out_idx, self.desc,
-                                              self.total_values)
+                                              self.total_values,
+                                              int(self.wire_half))
 
     def decode_sum(self, wires2d):
         from deepreduce_amd import _hip_ops
 
         return _hip_ops.batched_decode_sum(wires2d, self.desc, self.b2t,
-                                           self.total_values, self.mask_words)
+                                           self.total_values, self.mask_words,
+                                           int(self.wire_half))
 
 
 class BothPipeline(BatchedPipeline):
@@ -317,6 +322,7 @@ def maybe_pipeline(communicator, comp, named_tensors):
         comp.sparsifier.compress_ratio,
         params.get("fpr"),
         params.get("poly_degree", 5),
+        params.get("wire_dtype"),
         str(grads[0].device),
     )
     cached = getattr(communicator, "_bt_pipeline", None)

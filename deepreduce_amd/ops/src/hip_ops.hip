@@ -17,6 +17,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
 
 #define WAVE 64
 #define CHECK_CUDA(x) TORCH_CHECK(x.is_cuda(), #x " must be a GPU tensor")
@@ -1217,6 +1218,7 @@ __global__ void bt_qscatter_own_kernel(const uint64_t* __restrict__ mask,
                                        const int* __restrict__ b2t,
                                        uint8_t* __restrict__ wire /*nullable*/,
                                        float* __restrict__ vals_out /*nullable*/,
+                                       int wire_half,
                                        int64_t* __restrict__ out_idx) {
     const int t = b2t[blockIdx.x];
     const int64_t* D = bt_row(desc, t);
@@ -1227,6 +1229,7 @@ __global__ void bt_qscatter_own_kernel(const uint64_t* __restrict__ mask,
     const int64_t mwoff = D[9];
     const float* __restrict__ v = vals ? vals + D[1] : nullptr;
     float* __restrict__ wv = wire ? (float*)(wire + D[7]) : nullptr;
+    __half* __restrict__ wh = wire ? (__half*)(wire + D[7]) : nullptr;
     float* __restrict__ vo = vals_out ? vals_out + D[3] : nullptr;
     int64_t* __restrict__ oi = out_idx + D[3];
     int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
@@ -1247,7 +1250,10 @@ __global__ void bt_qscatter_own_kernel(const uint64_t* __restrict__ mask,
             int64_t ord = base_s + wbase + __popcll(ball & below);
             if (ord < k) {
                 oi[ord] = i;
-                if (wv) wv[ord] = v[i];
+                if (wv) {
+                    if (wire_half) wh[ord] = __float2half(v[i]);
+                    else wv[ord] = v[i];
+                }
                 if (vo) vo[ord] = v[i];
             }
         }
@@ -1268,7 +1274,7 @@ __global__ void bt_qscatter_add_kernel(const uint64_t* __restrict__ mask,
                                        const int* __restrict__ qoffs, int r,
                                        int64_t BV, int64_t MW,
                                        const uint8_t* __restrict__ wires,
-                                       int64_t wstride,
+                                       int64_t wstride, int wire_half,
                                        const int64_t* __restrict__ desc,
                                        const int* __restrict__ b2t,
                                        float* __restrict__ dense) {
@@ -1279,7 +1285,9 @@ __global__ void bt_qscatter_add_kernel(const uint64_t* __restrict__ mask,
     const int64_t end = min(start + BT_CHUNK, D[0]);
     const int64_t k = D[2];
     const int64_t mwoff = D[9];
-    const float* __restrict__ wv = (const float*)(wires + r * wstride + D[7]);
+    const uint8_t* __restrict__ wbase = wires + r * wstride + D[7];
+    const float* __restrict__ wv = (const float*)wbase;
+    const __half* __restrict__ wh = (const __half*)wbase;
     float* __restrict__ dv = dense + D[1];
     int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
     __shared__ int wave_cnt[QBLOCK / WAVE];
@@ -1297,7 +1305,7 @@ __global__ void bt_qscatter_add_kernel(const uint64_t* __restrict__ mask,
             for (int w = 0; w < wid; ++w) wbase += wave_cnt[w];
             uint64_t below = (lane == 63) ? (~0ull >> 1) : ((1ull << lane) - 1);
             int64_t ord = base_s + wbase + __popcll(ball & below);
-            if (ord < k) dv[i] += wv[ord];
+            if (ord < k) dv[i] += wire_half ? __half2float(wh[ord]) : wv[ord];
         }
         __syncthreads();
         if (threadIdx.x == 0) {
@@ -1313,7 +1321,7 @@ __global__ void bt_qscatter_add_kernel(const uint64_t* __restrict__ mask,
 __global__ void bt_scatter_dense_kernel(const int64_t* __restrict__ out_idx,
                                         const uint8_t* __restrict__ wire,
                                         const int64_t* __restrict__ desc,
-                                        int nT, int64_t K,
+                                        int nT, int64_t K, int wire_half,
                                         float* __restrict__ dense) {
     int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -1324,8 +1332,10 @@ __global__ void bt_scatter_dense_kernel(const int64_t* __restrict__ out_idx,
             if (bt_row(desc, mid)[3] <= j) lo = mid; else hi = mid - 1;
         }
         const int64_t* D = bt_row(desc, lo);
-        const float* wv = (const float*)(wire + D[7]);
-        dense[D[1] + out_idx[j]] = wv[j - D[3]];
+        const int64_t local = j - D[3];
+        dense[D[1] + out_idx[j]] = wire_half
+            ? __half2float(((const __half*)(wire + D[7]))[local])
+            : ((const float*)(wire + D[7]))[local];
     }
 }
 
@@ -1350,7 +1360,8 @@ std::vector<torch::Tensor> batched_compress(torch::Tensor values_flat,
                                             torch::Tensor desc,
                                             torch::Tensor b2t,
                                             int64_t wire_bytes, int64_t k_total,
-                                            int64_t mask_words) {
+                                            int64_t mask_words,
+                                            int64_t wire_half) {
     CHECK_CUDA(values_flat);
     auto v = values_flat.contiguous();
     auto d = desc.contiguous();
@@ -1403,14 +1414,15 @@ std::vector<torch::Tensor> batched_compress(torch::Tensor values_flat,
                        qcounts, dp, T, BV, qoffs);
     hipLaunchKernelGGL(bt_qscatter_own_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
                        (const uint64_t*)mask.data_ptr<int64_t>(), qoffs, vp, dp, mp,
-                       wire.data_ptr<uint8_t>(), (float*)nullptr,
+                       wire.data_ptr<uint8_t>(), (float*)nullptr, (int)wire_half,
                        out_idx.data_ptr<int64_t>());
     return {wire, out_idx};
 }
 
 // Own-payload decode: (wire, out_idx) -> dense flat
 torch::Tensor batched_scatter_dense(torch::Tensor wire, torch::Tensor out_idx,
-                                    torch::Tensor desc, int64_t total_values) {
+                                    torch::Tensor desc, int64_t total_values,
+                                    int64_t wire_half) {
     CHECK_CUDA(wire);
     auto d = desc.contiguous();
     const int T = (int)d.size(0);
@@ -1422,7 +1434,8 @@ torch::Tensor batched_scatter_dense(torch::Tensor wire, torch::Tensor out_idx,
                        dense.data_ptr<float>(), total_values);
     hipLaunchKernelGGL(bt_scatter_dense_kernel, dim3(bt_grid(K)), dim3(256), 0, stream,
                        out_idx.data_ptr<int64_t>(), wire.data_ptr<uint8_t>(),
-                       d.data_ptr<int64_t>(), T, K, dense.data_ptr<float>());
+                       d.data_ptr<int64_t>(), T, K, (int)wire_half,
+                       dense.data_ptr<float>());
     return dense;
 }
 
@@ -1431,7 +1444,7 @@ torch::Tensor batched_scatter_dense(torch::Tensor wire, torch::Tensor out_idx,
 // sequentially so the accumulation order is deterministic on every rank.
 torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
                                  torch::Tensor b2t, int64_t total_values,
-                                 int64_t mask_words) {
+                                 int64_t mask_words, int64_t wire_half) {
     CHECK_CUDA(wires2d);
     TORCH_CHECK(wires2d.dim() == 2, "expected [R, W]");
     auto w = wires2d.contiguous();
@@ -1462,7 +1475,7 @@ torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
     for (int r = 0; r < R; ++r) {
         hipLaunchKernelGGL(bt_qscatter_add_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
                            (const uint64_t*)mask.data_ptr<int64_t>(), qoffs, r, BV,
-                           mask_words, w.data_ptr<uint8_t>(), W,
+                           mask_words, w.data_ptr<uint8_t>(), W, (int)wire_half,
                            d.data_ptr<int64_t>(), map.data_ptr<int>(),
                            dense.data_ptr<float>());
     }
@@ -2076,7 +2089,7 @@ std::vector<torch::Tensor> batched_compress_both(
     // FP-aware gather to the temp buffer; positives into out_idx
     hipLaunchKernelGGL(bt_qscatter_own_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
                        (const uint64_t*)mask.data_ptr<int64_t>(), qoffs, vp, dp, mp,
-                       (uint8_t*)nullptr, vals_tmp.data_ptr<float>(),
+                       (uint8_t*)nullptr, vals_tmp.data_ptr<float>(), 0,
                        out_idx.data_ptr<int64_t>());
     // padded descending sort (stable: deterministic tie order)
     hipLaunchKernelGGL(bt2_fill_ninf, dim3(bt_grid((int64_t)T * kmax)), dim3(256), 0,
@@ -2156,7 +2169,7 @@ torch::Tensor batched_decode_both_sum(torch::Tensor wires2d, torch::Tensor desc,
                            (const uint64_t*)mask.data_ptr<int64_t>() + (int64_t)r * mask_words,
                            qoffs + (int64_t)r * BV, (const float*)nullptr,
                            d.data_ptr<int64_t>(), map.data_ptr<int>(),
-                           (uint8_t*)nullptr, (float*)nullptr,
+                           (uint8_t*)nullptr, (float*)nullptr, 0,
                            positives.data_ptr<int64_t>());
         hipLaunchKernelGGL(bt2_starts_from_wire_kernel, dim3(T), dim3(WAVE), 0, stream,
                            wr, d.data_ptr<int64_t>(), d1, starts.data_ptr<int64_t>());

@@ -181,8 +181,8 @@ class IndexCompressor(_WrapperBase):
         bits = torch.stack([p[1].contiguous() for p in payloads])
         # sync-free: [R, k] leftmost positives, one fused scatter-add
         idxs = ops.bloom_query_leftmost(bits, m, num_hash, numel, num_indices)
-        vals = torch.stack([p[0] for p in payloads])
-        dense = torch.zeros(numel, dtype=vals0.dtype, device=vals0.device)
+        vals = torch.stack([p[0] for p in payloads]).float()  # fp16 wire ok
+        dense = torch.zeros(numel, dtype=torch.float32, device=vals0.device)
         dense.index_add_(0, idxs.reshape(-1), vals.reshape(-1))
         return dense.view(shape)
 

@@ -156,3 +156,29 @@ def test_query_multi_cpu_fallback(data):
     assert counts.tolist() == [p1.numel(), p2.numel()]
     assert torch.equal(pos[: p1.numel()], p1)
     assert torch.equal(pos[p1.numel() :], p2)
+
+
+def test_fp16_wire_values_roundtrip():
+    """wire_dtype=fp16 halves the value bytes; decode returns float32 with
+    half-precision error, and the own-decode cache matches the wire decode
+    (so the residual absorbs the quantization)."""
+    import torch
+
+    from deepreduce_amd.codecs import compressor
+
+    torch.manual_seed(0)
+    t = torch.randn(50_000)
+    k = 500
+    vals, idxs = t.abs().topk(k)
+    vals = t[idxs]
+    params = {"policy": "leftmost", "dense_tensor": t, "wire_dtype": "fp16"}
+    v, bits, shape = compressor["bloom"].compress((vals, idxs, t.size()), params)
+    assert v.dtype == torch.float16
+    own = params.pop("_own_decoded")
+    v2, i2, _ = compressor["bloom"].decompress((v, bits, shape),
+                                               {"policy": "leftmost",
+                                                "wire_dtype": "fp16"})
+    assert v2.dtype == torch.float32
+    assert torch.equal(own[1], i2)
+    assert torch.equal(own[0], v2)  # cache == wire decode (residual exact)
+    assert torch.allclose(v2, t[i2], rtol=1e-3, atol=1e-4)

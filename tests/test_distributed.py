@@ -335,3 +335,28 @@ def test_overlapped_reducer_matches_sync_optimizer():
         p.join(timeout=60)
     for rank, ok in results:
         assert ok is True, f"rank {rank}: {ok}"
+
+
+def test_bench_driver_contract_2proc_cpu():
+    """Run bench.py exactly as the driver does (torch.distributed.run,
+    nproc=2), on CPU/gloo: the full integration path incl. rank-0 JSON."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+        "--master-port", "29770", os.path.join(repo, "bench.py"),
+        "--gpus", "2", "--steps", "2", "--warmup", "1", "--batch", "4",
+        "--model", "resnet20", "--device", "cpu",
+    ]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                         cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2 and d["metric"] == "images/sec"
+    assert d["config"]["parallelism"] == "dp2"
+    assert 0 < d["config"]["rel_volume"] < 0.2

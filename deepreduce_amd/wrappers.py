@@ -59,13 +59,12 @@ class _WrapperBase(Compressor):
 class ValueCompressor(_WrapperBase):
     """Sparsify, then post-compress the values array.
 
-    Parity: pytorch/deepreduce.py:51-97.  polyfit keeps uniform payload
-    sizes ragged-free only with the 'both'-style segment count — per-rank
-    num_pos changes segment counts, so like the reference (:364-367 note)
-    any value codec other than 'polyfit' marks payloads ragged; polyfit
-    itself is ragged too whenever ranks disagree on segment count, so we
-    conservatively mark ragged unless the sparsifier says otherwise AND the
-    codec is order/size stable ('qsgd').
+    Parity: pytorch/deepreduce.py:51-97.  Payload sizes: qsgd/polyseg
+    depend only on k and polyfit's padded slot layout only on N (see
+    codecs/polyfit.py get_segments), so those three gather uniformly;
+    other value codecs (gzip byte streams, doubleexp is uniform but kept
+    conservative) mark payloads ragged like the reference's note at
+    pytorch/deepreduce.py:364-367.
     """
 
     def __init__(self, sparsifier, params=None):

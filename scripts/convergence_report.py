@@ -27,6 +27,8 @@ CONFIGS = [
     ("Top-r 1%", {}),
     ("DR-BF leftmost", {"deepreduce": "index", "index": "bloom",
                         "policy": "leftmost"}),
+    ("DR-BF fp16 wire", {"deepreduce": "index", "index": "bloom",
+                         "policy": "leftmost", "wire_dtype": "fp16"}),
     ("DR-BF-P0", {"deepreduce": "index", "index": "bloom", "policy": "p0"}),
     ("DR-FitPoly", {"deepreduce": "value", "value": "polyfit"}),
     ("DR-QSGD", {"deepreduce": "value", "value": "qsgd"}),

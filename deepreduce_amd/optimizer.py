@@ -69,10 +69,16 @@ def _graph_safe(grc) -> bool:
             return False
     comp = getattr(grc, "compressor", None)
     params = getattr(comp, "params", None)
-    if params is None:  # bare sparsifier
-        from .compressors import ThresholdCompressor
+    from .compressors import RandomKCompressor, ThresholdCompressor
 
-        return not isinstance(comp, ThresholdCompressor)
+    if params is None:  # bare sparsifier
+        return not isinstance(comp, (ThresholdCompressor, RandomKCompressor))
+    # threshold: data-dependent payload sizes would replay with stale
+    # shapes; randomk: the per-step CPU reseed would freeze at the
+    # captured step — both eager only
+    if isinstance(getattr(comp, "sparsifier", None),
+                  (ThresholdCompressor, RandomKCompressor)):
+        return False
     mode = params.get("deepreduce")
     if not mode:
         return True

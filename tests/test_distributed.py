@@ -181,8 +181,13 @@ def _run_fused_equivalence(rank, world, mode, q):
         }
         if mode == "index":
             params.update({"deepreduce": "index", "index": "bloom", "policy": "leftmost"})
+        elif mode == "fp16":
+            params.update({"deepreduce": "index", "index": "bloom",
+                           "policy": "leftmost", "wire_dtype": "fp16"})
         elif mode == "both":
             params.update({"deepreduce": "both", "index": "bloom", "value": "polyfit"})
+        elif mode == "value":
+            params.update({"deepreduce": "value", "value": "polyfit"})
         elif mode == "dense":
             params = {"compressor": "none", "memory": "none", "communicator": "allreduce"}
         grc_a = deepreduce_from_params(dict(params))
@@ -205,7 +210,7 @@ def _run_fused_equivalence(rank, world, mode, q):
         q.put((rank, f"ERROR: {e!r}"))
 
 
-@pytest.mark.parametrize("mode", ["plain", "index", "both", "dense"])
+@pytest.mark.parametrize("mode", ["plain", "index", "fp16", "both", "value", "dense"])
 def test_fused_step_many_matches_per_tensor(mode):
     world = 2
     ctx = mp.get_context("spawn")

@@ -33,6 +33,9 @@ _SPARSIFIER_KWARGS = {
 
 
 def grace_from_params(params: dict):
+    from .params import validate
+
+    validate(params)
     comp_name = params.get("compressor", "topk")
     mem_name = params.get("memory", "none")
     comm_name = params.get("communicator", "allgather")

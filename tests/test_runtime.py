@@ -256,3 +256,18 @@ def test_model_zoo_forward_backward():
         loss.backward()
         assert all(p.grad is not None for p in model.parameters()
                    if p.requires_grad)
+
+
+def test_params_validation():
+    import pytest as _pytest
+
+    from deepreduce_amd import deepreduce_from_params
+
+    with _pytest.raises(ValueError, match="unknown params key"):
+        deepreduce_from_params({"compresor": "topk"})  # typo
+    with _pytest.raises(ValueError, match="not in"):
+        deepreduce_from_params({"deepreduce": "bot"})
+    with _pytest.raises(ValueError, match="compress_ratio"):
+        deepreduce_from_params({"compress_ratio": 3.0})
+    # reference-compat key accepted and ignored
+    deepreduce_from_params({"compressor": "topk", "hash_table": "/tmp/x.pt"})

@@ -99,40 +99,54 @@ class ResidualMemory(Memory):
             torch._foreach_add_(out, rs, alpha=self.beta)
             return out
 
-        if getattr(self, "_flat_key", None) != key:
+        pools = getattr(self, "_pools", None)
+        if pools is None:
+            pools = self._pools = {}
+        pool = pools.get(key)
+        if pool is None:
             total = sum(t.numel() for t in tensors)
-            self._flat_r = torch.zeros(total, dtype=tensors[0].dtype,
-                                       device=tensors[0].device)
-            self._flat_key = key
-            self._flat_offsets = []
+            flat_r = torch.zeros(total, dtype=tensors[0].dtype,
+                                 device=tensors[0].device)
+            offsets = []
             off = 0
             for n, t in zip(names, tensors):
-                self._flat_offsets.append(off)
+                offsets.append(off)
                 old = self.residuals.get(n)
-                view = self._flat_r[off : off + t.numel()].view(t.shape)
+                view = flat_r[off : off + t.numel()].view(t.shape)
                 if old is not None and old.shape == t.shape:
-                    view.copy_(old.to(view.device, view.dtype))  # checkpoint resume
+                    view.copy_(old.to(view.device, view.dtype))  # ckpt resume
                 # expose per-name views so state_dict/checkpoint still works
                 self.residuals[n] = view
                 off += t.numel()
+            pool = pools[key] = {"r": flat_r, "offsets": offsets}
+        flat_r = pool["r"]
         g_flat = torch.cat([t.reshape(-1) for t in tensors])
         if self.beta == 1.0 and self.gamma == 1.0:
-            c_flat = g_flat.add_(self._flat_r)  # g_flat is a fresh buffer
+            c_flat = g_flat.add_(flat_r)  # g_flat is a fresh buffer
         else:
-            c_flat = g_flat.mul_(self.gamma).add_(self._flat_r, alpha=self.beta)
-        self._flat_c = c_flat
+            c_flat = g_flat.mul_(self.gamma).add_(flat_r, alpha=self.beta)
+        pool["c"] = c_flat
+        self._flat_c = c_flat  # most-recent pool (fast-path identity check)
+        self._flat_r = flat_r
+        self._flat_key = key
         out = []
-        for off, t in zip(self._flat_offsets, tensors):
+        for off, t in zip(pool["offsets"], tensors):
             out.append(c_flat[off : off + t.numel()].view(t.shape))
         return out
 
     def update_many(self, tensors, names, decompressed):
         """residual <- compensated - decompressed, in place (graph-stable)."""
-        if getattr(self, "_flat_key", None) is not None and len(tensors) > 1 \
-                and tensors[0].data_ptr() == self._flat_c.data_ptr():
-            d_flat = torch.cat([d.reshape(-1) for d in decompressed])
-            torch.sub(self._flat_c, d_flat, out=self._flat_r)
-            return
+        pools = getattr(self, "_pools", None)
+        if pools and len(tensors) > 1:
+            total = sum(t.numel() for t in tensors)
+            for pool in pools.values():
+                c = pool.get("c")
+                if (c is not None and c.numel() == total
+                        and tensors[0].data_ptr() == c.data_ptr()):
+                    d_flat = torch.cat([d.reshape(-1) for d in decompressed])
+                    torch.sub(c, d_flat, out=pool["r"])
+                    pool["c"] = None  # consumed: avoid stale-aliasing matches
+                    return
         rs = [self.residuals[n] for n in names]
         torch._foreach_copy_(rs, list(tensors))
         torch._foreach_sub_(rs, [d.view_as(t) for d, t in zip(decompressed, tensors)])
@@ -146,7 +160,8 @@ class ResidualMemory(Memory):
         self.residuals = state["residuals"]
         self.beta = state["beta"]
         self.gamma = state["gamma"]
-        self._flat_key = None  # rebuild the flat pool from loaded values
+        self._flat_key = None  # rebuild the flat pools from loaded values
+        self._pools = {}
 
 
 memory_registry = {

@@ -325,10 +325,16 @@ def maybe_pipeline(communicator, comp, named_tensors):
         params.get("wire_dtype"),
         str(grads[0].device),
     )
-    cached = getattr(communicator, "_bt_pipeline", None)
-    if cached is not None and cached[0] == key:
-        return cached[1]
-    bp = cls([n for n, _ in named_tensors],
-             [t.numel() for t in grads], params, grads[0].device)
-    communicator._bt_pipeline = (key, bp)
+    cache = getattr(communicator, "_bt_pipelines", None)
+    if cache is None:
+        cache = communicator._bt_pipelines = {}
+    bp = cache.get(key)
+    if bp is None:
+        bp = cls([n for n, _ in named_tensors],
+                 [t.numel() for t in grads], params, grads[0].device)
+        cache[key] = bp
+        # legacy single-slot attribute (tests introspect it)
+        communicator._bt_pipeline = (key, bp)
+    else:
+        communicator._bt_pipeline = (key, bp)
     return bp

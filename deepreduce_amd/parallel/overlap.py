@@ -1,15 +1,24 @@
-"""Overlapped, bucketed gradient exchange (MI355X/xGMI-first).
+"""Bucketed gradient exchange overlapped with backward (MI355X/xGMI-first).
 
-Design (SURVEY.md sect. 2.3 collective notes): compressed payloads are 1-2%
-of the gradient, so per-tensor collective LATENCY dominates on xGMI — the
-wins are (a) fusing the ~100 tiny tensors (BatchNorm etc., <= the wrapper's
-1000-element bypass) into ONE dense flat all-reduce instead of one ragged
-allgather each, and (b) running compression + collectives of large tensors
-on a side HIP stream as soon as each gradient is produced by backward
-(post-accumulate-grad hooks), overlapping with the rest of backward.
+Design (SURVEY.md §2.3 collective notes, BASELINE north star "decompress
+overlapped with backward on a side HIP stream"): the model's large tensors
+are partitioned — in reverse registration order, i.e. the order backward
+produces gradients — into a few buckets of roughly equal element count.
+As soon as every gradient of a bucket has been produced
+(post-accumulate-grad hooks), the bucket's whole compression pipeline
+(compensate → batched compress → residual update → collective) is enqueued
+on a side HIP stream, overlapping the rest of backward. `finalize()`
+drains the stream, decodes each bucket (multi-rank-batched), writes the
+reduced gradients, and runs the fused dense exchange of the small
+(≤1000-element) tensors.
 
-The residual-update ordering invariant (compensate -> compress -> update,
-SURVEY.md sect. 7) is preserved per tensor inside the hook.
+Each bucket gets its own batched pipeline (ops/batched.py) and flat
+residual pool; on CPU (or for non-qualifying codecs) the bucket falls back
+to the communicator's generic step_many semantics — bit-identical to the
+synchronous DistributedOptimizer path (tested on gloo world=2).
+
+The residual-update ordering invariant (compensate → compress → update,
+SURVEY.md §7) is preserved per bucket inside the hook.
 """
 from __future__ import annotations
 
@@ -20,146 +29,146 @@ __all__ = ["OverlappedReducer"]
 
 
 class OverlappedReducer:
-    """Gradient reducer with hook-driven compression and bucketed small
-    tensors.  Usage:
+    """Hook-driven bucketed reducer.  Usage:
 
-        reducer = OverlappedReducer(model, grc, small_threshold=1000)
+        reducer = OverlappedReducer(model, grc, num_buckets=3)
         ...
-        loss.backward()          # hooks fire as grads are produced
-        reducer.finalize()       # drain comm stream, write reduced grads
+        loss.backward()          # buckets launch as their grads complete
+        reducer.finalize()       # drain, decode, write reduced grads
         optimizer.step()
     """
 
-    def __init__(self, model: torch.nn.Module, grc, small_threshold: int = 1000):
+    def __init__(self, model: torch.nn.Module, grc, num_buckets: int = 3,
+                 small_threshold: int = 1000):
         self.model = model
         self.grc = grc
         self.small_threshold = small_threshold
         self.last_wire_bytes = 0
 
-        self._large: list[tuple[str, torch.nn.Parameter]] = []
-        self._small: list[tuple[str, torch.nn.Parameter]] = []
-        for name, p in model.named_parameters():
-            if not p.requires_grad:
-                continue
-            (self._large if p.numel() > small_threshold else self._small).append((name, p))
+        params = [(n, p) for n, p in model.named_parameters() if p.requires_grad]
+        self._small = [(n, p) for n, p in params if p.numel() <= small_threshold]
+        large = [(n, p) for n, p in reversed(params) if p.numel() > small_threshold]
 
-        self._use_cuda = any(p.is_cuda for _, p in self._large + self._small)
+        # equal-elements partition in backward production order
+        total = sum(p.numel() for _, p in large)
+        num_buckets = max(1, min(num_buckets, len(large)))
+        target = total / num_buckets
+        self._buckets: list[list] = [[]]
+        acc = 0
+        for n, p in large:
+            if acc >= target * len(self._buckets) and len(self._buckets) < num_buckets:
+                self._buckets.append([])
+            self._buckets[-1].append((n, p))
+            acc += p.numel()
+        self._bucket_of = {n: b for b, bucket in enumerate(self._buckets)
+                           for n, _ in bucket}
+
+        self._use_cuda = any(p.is_cuda for _, p in params)
         self._comm_stream = torch.cuda.Stream() if self._use_cuda else None
-        self._pending: dict[str, tuple] = {}
+        self._arrived = [0] * len(self._buckets)
+        self._launched: dict[int, tuple] = {}
         self._hooks = []
-        self._install_hooks()
-
-        # persistent fused buffer for the small-tensor dense path
-        n_small = sum(p.numel() for _, p in self._small)
-        dev = self._small[0][1].device if self._small else torch.device("cpu")
-        self._small_buf = torch.zeros(n_small, dtype=torch.float32, device=dev)
-
-    # -- hooks ------------------------------------------------------------
-    def _install_hooks(self):
-        for name, p in self._large:
+        for n, p in large:
             self._hooks.append(
-                p.register_post_accumulate_grad_hook(self._make_hook(name, p))
-            )
+                p.register_post_accumulate_grad_hook(self._make_hook(n)))
 
-    def _make_hook(self, name, p):
-        def hook(_param):
-            self._start_exchange(name, p)
-
-        return hook
+        n_small = sum(p.numel() for _, p in self._small)
+        dev = params[0][1].device if params else torch.device("cpu")
+        self._small_buf = torch.zeros(n_small, dtype=torch.float32, device=dev)
 
     def remove_hooks(self):
         for h in self._hooks:
             h.remove()
         self._hooks = []
 
-    # -- per-tensor pipeline ----------------------------------------------
-    def _start_exchange(self, name, p):
-        grad = p.grad
-        if grad is None:
-            return
+    # -- hooks -------------------------------------------------------------
+    def _make_hook(self, name):
+        b = self._bucket_of[name]
+
+        def hook(_param):
+            self._arrived[b] += 1
+            if self._arrived[b] == len(self._buckets[b]):
+                self._arrived[b] = 0
+                self._launch(b)
+
+        return hook
+
+    # -- per-bucket pipeline -------------------------------------------------
+    def _launch(self, b):
+        grc = self.grc
+        named = [(n, p.grad.data.float()) for n, p in self._buckets[b]]
         if self._comm_stream is not None:
             self._comm_stream.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(self._comm_stream):
-                self._compress_and_launch(name, p, grad)
+            ctx = torch.cuda.stream(self._comm_stream)
         else:
-            self._compress_and_launch(name, p, grad)
+            ctx = _null_ctx()
+        with ctx:
+            from ..ops import batched as _bt
 
-    def _compress_and_launch(self, name, p, grad):
-        grc = self.grc
-        g32 = grad.data.float()
-        tensor = grc.memory.compensate(g32, name)
-        payload, ctx = grc.compressor.compress(tensor, name)
-        grc.memory.update(tensor, name, grc.compressor, payload, ctx)
-        work, gathered, metas = self._launch_collective(payload)
-        self._pending[name] = (p, payload, ctx, work, gathered, metas)
+            comp = grc.compressor
+            bp = _bt.maybe_pipeline(grc, comp, named)
+            if bp is None:
+                # generic fallback: synchronous bucket exchange now (still
+                # on the comm stream); result ready at finalize
+                impl = getattr(grc, "_step_many_impl", None)
+                outs = impl(named) if impl else [grc.step(t, n) for n, t in named]
+                self._launched[b] = ("done", outs, grc.last_wire_bytes)
+                return
+            names = [n for n, _ in named]
+            grads = [t for _, t in named]
+            compensated = grc.memory.compensate_many(grads, names)
+            flat_c = getattr(grc.memory, "_flat_c", None)
+            if (flat_c is not None and compensated
+                    and compensated[0].data_ptr() == flat_c.data_ptr()
+                    and flat_c.numel() == bp.total_values):
+                c_flat = flat_c
+            else:
+                c_flat = torch.cat([t.reshape(-1) for t in compensated])
+            wire, own = bp.compress_and_own(c_flat)
+            torch.sub(c_flat, own, out=grc.memory._flat_r)
+            world = grc.world_size
+            if world == 1:
+                self._launched[b] = ("own", bp, own, wire.numel())
+                return
+            gathered = torch.empty(world, wire.numel(), dtype=torch.uint8,
+                                   device=wire.device)
+            try:
+                work = dist.all_gather_into_tensor(gathered.view(-1), wire,
+                                                   async_op=True)
+            except (AttributeError, RuntimeError):
+                bufs = list(gathered.unbind(0))
+                work = dist.all_gather(bufs, wire, async_op=True)
+            self._launched[b] = ("gathered", bp, gathered, work, wire.numel())
 
-    def _launch_collective(self, payload):
-        from ..communicator import _flatten_payload
-
-        world = dist.get_world_size() if dist.is_available() and dist.is_initialized() else 1
-        buffer, metas = _flatten_payload(payload)
-        self.last_wire_bytes += buffer.numel()
-        if world == 1:
-            return None, None, metas
-        if self.grc.compressor.tensors_size_are_same:
-            gathered = [torch.empty_like(buffer) for _ in range(world)]
-            work = dist.all_gather(gathered, buffer, async_op=True)
-            return work, gathered, metas
-        # ragged: sizes first (sync, tiny), then padded async gather
-        counts = torch.tensor([buffer.numel()], dtype=torch.int64, device=buffer.device)
-        all_counts = [torch.empty_like(counts) for _ in range(world)]
-        dist.all_gather(all_counts, counts)
-        max_bytes = max(int(c.item()) for c in all_counts)
-        padded = torch.zeros(max_bytes, dtype=torch.uint8, device=buffer.device)
-        padded[: buffer.numel()] = buffer
-        gathered = [torch.empty_like(padded) for _ in range(world)]
-        work = dist.all_gather(gathered, padded, async_op=True)
-        return work, gathered, (metas, [int(c.item()) for c in all_counts])
-
-    # -- finalize ----------------------------------------------------------
+    # -- finalize ------------------------------------------------------------
     def finalize(self):
-        """Drain: complete collectives, decompress+average, write grads;
-        then the fused dense exchange of the small tensors."""
-        from ..communicator import _unflatten_payload
-
-        world = dist.get_world_size() if dist.is_available() and dist.is_initialized() else 1
-        stream_ctx = (
-            torch.cuda.stream(self._comm_stream) if self._comm_stream is not None else _null_ctx()
-        )
-        with stream_ctx:
-            for name, p in self._large:
-                if name not in self._pending:
-                    # hook didn't fire (grad absent) — skip
-                    continue
-                p_, payload, ctx, work, gathered, metas = self._pending.pop(name)
-                grc = self.grc
-                if work is None:
-                    out = grc.compressor.decompress(payload, ctx)
+        grc = self.grc
+        world = grc.world_size
+        ctx = (torch.cuda.stream(self._comm_stream)
+               if self._comm_stream is not None else _null_ctx())
+        with ctx:
+            for b, bucket in enumerate(self._buckets):
+                entry = self._launched.pop(b, None)
+                if entry is None:
+                    continue  # no grads this step
+                if entry[0] == "done":
+                    _, outs, wb = entry
+                    self.last_wire_bytes += wb
+                elif entry[0] == "own":
+                    _, bp, own, wb = entry
+                    self.last_wire_bytes += wb
+                    outs = _split(own, [p for _, p in bucket])
                 else:
-                    work.wait()
-                    if isinstance(metas, tuple):  # ragged
-                        entry_metas, counts = metas
-                        payloads = []
-                        for r in range(world):
-                            nb = counts[r]
-                            # per-entry numels unknown for ragged fused wire;
-                            # fall back: exchange already done via sizes of
-                            # full buffer only works for single-entry ragged.
-                            payloads.append(
-                                _unflatten_ragged(gathered[r][:nb], entry_metas)
-                            )
-                    else:
-                        payloads = [_unflatten_payload(b, metas) for b in gathered]
-                    total = None
-                    batch = getattr(grc.compressor, "decompress_batch", None)
-                    if batch is not None:
-                        total = batch(payloads, ctx)
-                    if total is None:
-                        for pay in payloads:
-                            d = grc.compressor.decompress(pay, ctx)
-                            total = d if total is None else total + d
-                    out = total / world if grc.compressor.average else total
-                p_.grad.data.copy_(out.view_as(p_.grad.data))
+                    _, bp, gathered, work, wb = entry
+                    self.last_wire_bytes += wb
+                    if work is not None:
+                        work.wait()
+                    result = bp.decode_sum(gathered)
+                    if grc.compressor.average:
+                        result /= world
+                    outs = _split(result, [p for _, p in bucket])
+                for (n, p), r in zip(bucket, outs):
+                    p.grad.data.copy_(r.view_as(p.grad.data))
             self._exchange_small(world)
         if self._comm_stream is not None:
             torch.cuda.current_stream().wait_stream(self._comm_stream)
@@ -181,31 +190,23 @@ class OverlappedReducer:
         for name, p in self._small:
             n = p.numel()
             if p.grad is not None:
-                p.grad.data.copy_(self._small_buf[offset : offset + n].view_as(p.grad.data))
+                p.grad.data.copy_(
+                    self._small_buf[offset : offset + n].view_as(p.grad.data))
             offset += n
 
     def zero_wire_counter(self):
         self.last_wire_bytes = 0
 
 
-def _unflatten_ragged(buffer, metas):
-    """Ragged payloads: only the TOTAL byte length varies per rank; the
-    entry structure is recovered by the compressor's decompress from the
-    wire itself.  We reconstruct entries proportionally is impossible in
-    general — instead ragged wires must be single-entry or self-describing.
-    For the codecs in this package the ragged cases (polyfit coeffs, p0
-    vals) are self-describing via headers, so we re-split using the header
-    conventions encoded in metas dtypes with trailing-entry absorption.
-    """
-    # single-entry fast path
-    if len(metas) == 1:
-        dtype, _ = metas[0]
-        esize = torch.empty(0, dtype=dtype).element_size()
-        return (buffer[: (buffer.numel() // esize) * esize].view(dtype),)
-    raise NotImplementedError(
-        "overlapped ragged multi-entry payloads: use the synchronous "
-        "Allgather communicator for this codec configuration"
-    )
+def _split(flat, params):
+    outs = []
+    off = 0
+    for p in params:
+        n = p.numel()
+        outs.append(flat[off : off + n].view(p.grad.shape
+                                             if p.grad is not None else p.shape))
+        off += n
+    return outs
 
 
 class _null_ctx:

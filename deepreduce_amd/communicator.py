@@ -87,6 +87,49 @@ class Communicator:
         SURVEY.md sect. 2.3 collective notes).  Default: per-tensor loop."""
         return [self.step(t, n) for n, t in named_tensors]
 
+    def _log_step(self, named_tensors):
+        """Per-step observability (reference parity: the C++ ops' per
+        rank/step/gradient stats files, compression_utils.hpp:96-176):
+        params['log_stats'] = directory enables a JSONL record per tensor
+        per step with wire/dense bytes; flushed every `frequency` (100)
+        steps and on interpreter exit."""
+        params = getattr(self, "params", {})
+        out = params.get("log_stats")
+        if not out:
+            return
+        if not hasattr(self, "_stats"):
+            import atexit
+
+            from .metrics import StatsLogger
+
+            self._stats = StatsLogger(out)
+            atexit.register(self._stats.dump)
+        st = self._stats
+        st.tick()
+        for n, t in named_tensors:
+            st.log(n, self._tensor_wire_bytes(n, t), t.numel() * t.element_size())
+        if st.step % 100 == 0:
+            st.dump()
+
+    def _tensor_wire_bytes(self, name, t):
+        """Static per-tensor wire estimate for logging (exact totals come
+        from last_wire_bytes)."""
+        comp = self.compressor
+        if t.numel() <= 1000 and getattr(self, "params", {}).get("small_dense", True):
+            return t.numel() * 4
+        metas = getattr(self, "_bt_pipeline", None)
+        if metas is not None:
+            bp = metas[1]
+            if name in bp.names:
+                i = bp.names.index(name)
+                return sum(
+                    n * torch.empty(0, dtype=d).element_size()
+                    for d, n in bp.metas[i]
+                )
+        ratio = getattr(getattr(comp, "sparsifier", comp), "compress_ratio", 0.01)
+        k = max(1, int(round(t.numel() * ratio)))
+        return 8 * k  # fallback rough estimate (fp32 vals + int32 idxs)
+
 
 class Allgather(Communicator):
     """All-gather of per-rank compressed payloads, local decompress, average.
@@ -147,6 +190,7 @@ class Allgather(Communicator):
         Disable with params['small_dense']=False for strict per-tensor
         reference semantics.
         """
+        self._log_step(named_tensors)
         small_dense = bool(getattr(self, "params", {}).get("small_dense", True))
         if small_dense:
             small = [(i, n, t) for i, (n, t) in enumerate(named_tensors)

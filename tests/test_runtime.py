@@ -271,3 +271,25 @@ def test_params_validation():
         deepreduce_from_params({"compress_ratio": 3.0})
     # reference-compat key accepted and ignored
     deepreduce_from_params({"compressor": "topk", "hash_table": "/tmp/x.pt"})
+
+
+def test_log_stats_wiring(tmp_path):
+    """params['log_stats'] produces per-step per-tensor JSONL records."""
+    import json
+
+    from deepreduce_amd import deepreduce_from_params
+
+    out = str(tmp_path / "stats")
+    grc = deepreduce_from_params({
+        "compressor": "topk", "memory": "residual",
+        "communicator": "allgather", "compress_ratio": 0.05,
+        "deepreduce": "index", "index": "bloom", "policy": "leftmost",
+        "log_stats": out,
+    })
+    for s in range(2):
+        grc.step_many([("w", torch.randn(5000)), ("b", torch.randn(64))])
+    fn = grc._stats.dump()
+    recs = [json.loads(l) for l in open(fn)]
+    assert len(recs) == 4  # 2 tensors x 2 steps
+    assert {r["tensor"] for r in recs} == {"w", "b"}
+    assert all(r["wire_bytes"] > 0 and r["dense_bytes"] > 0 for r in recs)

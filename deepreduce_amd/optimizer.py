@@ -131,9 +131,15 @@ class DistributedOptimizer:
             self.last_wire_bytes = reduce_gradients(self.model, self.grc)
             return
         if self._graph is not None:
-            self._graph.replay()
-            self.last_wire_bytes = self._graph_wire_bytes
-            return
+            if self._grad_ptrs() != self._graph_grad_ptrs:
+                # gradient storages changed (e.g. zero_grad(set_to_none=True)
+                # reallocated them): the captured pointers are stale —
+                # invalidate and re-capture below
+                self._graph = None
+            else:
+                self._graph.replay()
+                self.last_wire_bytes = self._graph_wire_bytes
+                return
         self._graph_calls += 1
         if self._graph_calls <= self._graph_warmup:
             self.last_wire_bytes = reduce_gradients(self.model, self.grc)
@@ -153,12 +159,17 @@ class DistributedOptimizer:
             with torch.cuda.graph(g):
                 self._graph_wire_bytes = reduce_gradients(self.model, self.grc)
             self._graph = g
+            self._graph_grad_ptrs = self._grad_ptrs()
             self.last_wire_bytes = self._graph_wire_bytes
         except Exception:
             self._use_graph = False
             self._graph = None
             torch.cuda.synchronize()
             self.last_wire_bytes = reduce_gradients(self.model, self.grc)
+
+    def _grad_ptrs(self):
+        return tuple(p.grad.data_ptr() for p in self.model.parameters()
+                     if p.grad is not None)
 
     def step(self, closure=None):
         self._exchange()
@@ -180,3 +191,4 @@ class DistributedOptimizer:
         mem = getattr(self.grc, "memory", None)
         if mem is not None and "memory" in sd:
             mem.load_state_dict(sd["memory"])
+        self._graph = None  # residual storages changed: re-capture

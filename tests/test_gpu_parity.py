@@ -600,3 +600,40 @@ def test_overlapped_reducer_gpu_matches_sync(dev):
     torch.cuda.synchronize()
     for a, b in zip(model_a.parameters(), model_b.parameters()):
         assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
+
+
+def test_graph_recaptures_on_grad_realloc(dev):
+    """zero_grad(set_to_none=True) reallocates gradient storages; the
+    captured exchange must detect the stale pointers and re-capture
+    instead of silently replaying old buffers."""
+    from deepreduce_amd import DistributedOptimizer, deepreduce_from_params
+
+    torch.manual_seed(41)
+    model = torch.nn.Sequential(torch.nn.Linear(2048, 512), torch.nn.ReLU(),
+                                torch.nn.Linear(512, 8)).to(dev)
+    grc = deepreduce_from_params({
+        "compressor": "topk", "memory": "residual",
+        "communicator": "allgather", "compress_ratio": 0.02,
+        "deepreduce": "index", "index": "bloom", "policy": "leftmost",
+    })
+    opt = DistributedOptimizer(torch.optim.SGD(model.parameters(), lr=0.05),
+                               grc, model, use_graph=True, graph_warmup=2)
+    gen = torch.Generator().manual_seed(5)
+
+    def one_step(set_to_none):
+        x = torch.randn(16, 2048, generator=gen).to(dev)
+        y = torch.randint(0, 8, (16,), generator=gen).to(dev)
+        opt.zero_grad(set_to_none=set_to_none)
+        torch.nn.functional.cross_entropy(model(x), y).backward()
+        opt.step()
+
+    for _ in range(4):
+        one_step(False)
+    assert opt._graph is not None
+    g1 = opt._graph
+    one_step(True)   # grads reallocated -> must recapture (not crash/corrupt)
+    one_step(True)
+    torch.cuda.synchronize()
+    assert opt._graph is not g1 or opt._graph is None
+    # training remains finite and sane
+    assert all(torch.isfinite(p).all() for p in model.parameters())

@@ -134,8 +134,15 @@ class DistributedOptimizer:
             if self._grad_ptrs() != self._graph_grad_ptrs:
                 # gradient storages changed (e.g. zero_grad(set_to_none=True)
                 # reallocated them): the captured pointers are stale —
-                # invalidate and re-capture below
+                # invalidate and re-capture below.  If it keeps happening
+                # (unstable grad storages every step), graphs can't help:
+                # stay eager.
                 self._graph = None
+                self._recaptures = getattr(self, "_recaptures", 0) + 1
+                if self._recaptures > 3:
+                    self._use_graph = False
+                    self.last_wire_bytes = reduce_gradients(self.model, self.grc)
+                    return
             else:
                 self._graph.replay()
                 self.last_wire_bytes = self._graph_wire_bytes

@@ -140,18 +140,29 @@ class Huffman(SparseCompressor):
             code += 1
             max_code_end[L] = code
 
-        out = np.empty(n, dtype=np.uint8)
-        pos = 0
-        for i in range(n):
-            acc = 0
-            L = 0
-            while True:
-                acc = (acc << 1) | int(bits[pos])
-                pos += 1
-                L += 1
-                if L in first_code and acc < max_code_end[L]:
-                    out[i] = sym_by_rank[first_rank[L] + (acc - first_code[L])]
-                    break
+        from .. import ops
+
+        if ops._cpu_native():
+            from deepreduce_amd import _hip_ops
+
+            out = _hip_ops.huffman_decode_cpu(
+                torch.from_numpy(stream.copy()), n,
+                torch.from_numpy(codes.astype(np.int64)),
+                torch.from_numpy(lengths.astype(np.int64)),
+            ).numpy()
+        else:
+            out = np.empty(n, dtype=np.uint8)
+            pos = 0
+            for i in range(n):
+                acc = 0
+                L = 0
+                while True:
+                    acc = (acc << 1) | int(bits[pos])
+                    pos += 1
+                    L += 1
+                    if L in first_code and acc < max_code_end[L]:
+                        out[i] = sym_by_rank[first_rank[L] + (acc - first_code[L])]
+                        break
         idxs_np = np.frombuffer(out.tobytes(), dtype=np.int32).copy() if n else np.empty(0, np.int32)
         idxs = torch.from_numpy(idxs_np).long().to(vals.device)
         return vals, idxs, shape

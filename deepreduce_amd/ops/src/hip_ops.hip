@@ -18,6 +18,8 @@
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
+#include <algorithm>
+#include <vector>
 
 #define WAVE 64
 #define CHECK_CUDA(x) TORCH_CHECK(x.is_cuda(), #x " must be a GPU tensor")
@@ -2443,6 +2445,54 @@ torch::Tensor pack_ints_cpu(torch::Tensor values, int64_t nbits) {
     return out;
 }
 
+
+torch::Tensor huffman_decode_cpu(torch::Tensor stream, int64_t n,
+                                 torch::Tensor codes, torch::Tensor lengths) {
+    // canonical Huffman decode of n byte-symbols (codecs/huffman.py tables):
+    // replaces the per-bit python loop (1.1 ms at n=1.5k) with ~ns/symbol
+    auto st = stream.contiguous();
+    auto cd = codes.to(torch::kInt64).contiguous();
+    auto ln = lengths.to(torch::kInt64).contiguous();
+    const uint8_t* bytes = st.data_ptr<uint8_t>();
+    const int64_t* C = cd.data_ptr<int64_t>();
+    const int64_t* L = ln.data_ptr<int64_t>();
+    // per-length tables
+    int64_t first_code[64], first_rank[64], max_end[64];
+    for (int i = 0; i < 64; ++i) { first_code[i] = -1; max_end[i] = -1; first_rank[i] = 0; }
+    // canonical order: (length, symbol)
+    std::vector<int> order;
+    for (int s2 = 0; s2 < 256; ++s2)
+        if (L[s2] > 0) order.push_back(s2);
+    std::stable_sort(order.begin(), order.end(),
+                     [&](int a, int b) { return L[a] != L[b] ? L[a] < L[b] : a < b; });
+    std::vector<uint8_t> sym_by_rank(order.begin(), order.end());
+    int64_t code = 0, prev_len = 0, rank = 0;
+    for (int s2 : order) {
+        int64_t len = L[s2];
+        code <<= (len - prev_len);
+        if (first_code[len] < 0) { first_code[len] = code; first_rank[len] = rank; }
+        prev_len = len;
+        max_end[len] = ++code;
+        ++rank;
+    }
+    auto out = torch::empty({n}, torch::dtype(torch::kUInt8));
+    uint8_t* o = out.data_ptr<uint8_t>();
+    int64_t pos = 0;  // bit position, LSB-first within each byte
+    for (int64_t i = 0; i < n; ++i) {
+        int64_t acc = 0, len = 0;
+        for (;;) {
+            acc = (acc << 1) | ((bytes[pos >> 3] >> (pos & 7)) & 1);
+            ++pos;
+            ++len;
+            if (first_code[len] >= 0 && acc < max_end[len]) {
+                o[i] = sym_by_rank[(size_t)(first_rank[len] + (acc - first_code[len]))];
+                break;
+            }
+        }
+    }
+    return out;
+}
+
 torch::Tensor unpack_ints_cpu(torch::Tensor stream, int64_t n, int64_t nbits) {
     auto s = stream.contiguous();
     const uint8_t* sv = s.data_ptr<uint8_t>();
@@ -2501,4 +2551,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");
     m.def("pack_ints_cpu", &pack_ints_cpu, "n-bit pack (C++ CPU)");
     m.def("unpack_ints_cpu", &unpack_ints_cpu, "n-bit unpack (C++ CPU)");
+    m.def("huffman_decode_cpu", &huffman_decode_cpu, "canonical Huffman decode (C++ CPU)");
 }

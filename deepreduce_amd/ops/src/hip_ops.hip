@@ -1479,7 +1479,9 @@ torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
 // normalized-x power sums and moments, reduce through wave+LDS, build the
 // ridged Gram and solve in-register Cholesky; thread 0 writes the d1
 // coefficients.  Used by the per-tensor and batched fit kernels.
+#define PF_SH_COLS (2 * PF_MAXD1 - 1 + PF_MAXD1)
 __device__ void pf_fit_segment(const float* __restrict__ y, int64_t len, int degree,
+                               double (*sh)[PF_SH_COLS] /*[QBLOCK/WAVE]*/,
                                double* __restrict__ out /*[d1]*/) {
     const int d1 = degree + 1;
     const int np = 2 * degree + 1;
@@ -1503,7 +1505,6 @@ __device__ void pf_fit_segment(const float* __restrict__ y, int64_t len, int deg
         for (int p = 0; p < np; ++p) ps[p] += __shfl_down(ps[p], off, WAVE);
         for (int p = 0; p < d1; ++p) mo[p] += __shfl_down(mo[p], off, WAVE);
     }
-    __shared__ double sh[QBLOCK / WAVE][2 * PF_MAXD1 - 1 + PF_MAXD1];
     if (lane == 0) {
         for (int p = 0; p < np; ++p) sh[wid][p] = ps[p];
         for (int p = 0; p < d1; ++p) sh[wid][np + p] = mo[p];
@@ -1554,7 +1555,9 @@ __global__ void polyfit_fit_kernel(const float* __restrict__ y,
     const int sgi = blockIdx.x;
     const int64_t start = seg_starts[sgi];
     const int64_t len = seg_starts[sgi + 1] - start;
-    pf_fit_segment(y + start, len, degree, coeffs + (int64_t)sgi * (degree + 1));
+    __shared__ double sh[QBLOCK / WAVE][PF_SH_COLS];
+    pf_fit_segment(y + start, len, degree, sh,
+                   coeffs + (int64_t)sgi * (degree + 1));
 }
 
 // Horner evaluation of the fitted piecewise polynomial (binary search for
@@ -1792,7 +1795,8 @@ __global__ void bt2_fit_kernel(const float* __restrict__ sorted /*[T,kmax]*/,
     const int64_t start = st[si];
     const int64_t len = st[si + 1] - start;
     double* c = (double*)(wire + D[12]) + (int64_t)si * (degree + 1);
-    pf_fit_segment(sorted + (int64_t)t * kmax + start, len, degree, c);
+    __shared__ double sh[QBLOCK / WAVE][PF_SH_COLS];
+    pf_fit_segment(sorted + (int64_t)t * kmax + start, len, degree, sh, c);
 }
 
 // pack each tensor's mapping (argsort of the padded sort) at nbits into the

@@ -609,49 +609,6 @@ __global__ void topk_hist2_kernel(const float* __restrict__ v, int64_t n,
         if (lh[b]) atomicAdd(&hist[b], lh[b]);
 }
 
-// Shared single-wave threshold derivation: bstar = largest bin with
-// suffix_sum(bin) >= target, above = suffix_sum(bstar+1).  Used by the
-// per-tensor and batched top-k kernels.
-__device__ __forceinline__ void tk_thresh_wave(const int* __restrict__ H, int target,
-                                               int* found_b, int* found_above) {
-    const int lane = threadIdx.x;
-    const int SEG = TK_BINS / WAVE;
-    int h[SEG];
-    int seg_sum = 0;
-    for (int j = 0; j < SEG; ++j) {
-        h[j] = H[lane * SEG + j];
-        seg_sum += h[j];
-    }
-    int suffix_excl;
-    {
-        int acc = seg_sum;
-        for (int off = 1; off < WAVE; off <<= 1) {
-            int other = __shfl_down(acc, off, WAVE);
-            if (lane + off < WAVE) acc += other;
-        }
-        suffix_excl = acc - seg_sum;
-    }
-    int fb = -1, fa = 0;
-    int acc = 0;
-    for (int j = SEG - 1; j >= 0; --j) {
-        int suffix_here = suffix_excl + acc + h[j];
-        if (suffix_here >= target) {
-            fb = lane * SEG + j;
-            fa = suffix_excl + acc;
-            break;
-        }
-        acc += h[j];
-    }
-    int best = fb;
-    for (int off = 1; off < WAVE; off <<= 1) {
-        int other = __shfl_down(best, off, WAVE);
-        if (lane + off < WAVE && other > best) best = other;
-    }
-    best = __shfl(best, 0, WAVE);
-    *found_b = (fb == best && fb >= 0) ? fb : -1;  // winner lane only
-    *found_above = fa;
-}
-
 // Derive the threshold bin from a 2048-bin histogram on-device (one
 // wavefront): bstar = largest bin b with suffix_sum(b) >= target, above =
 // suffix_sum(bstar+1).  Replaces the flip/cumsum/cat/index torch-op chain
@@ -661,10 +618,46 @@ __device__ __forceinline__ void tk_thresh_wave(const int* __restrict__ H, int ta
 // thresh22 = bstar1*2048 + bstar2 and count_above = above1 + above2.
 __global__ void topk_thresh_kernel(const int* __restrict__ hist, int64_t k, int level,
                                    int* __restrict__ sc) {
+    const int lane = threadIdx.x;               // single wave of 64
     const int target = (level == 1) ? (int)k : (int)k - sc[1];
-    int found_b, found_above;
-    tk_thresh_wave(hist, target, &found_b, &found_above);
-    if (found_b >= 0) {  // winner lane
+    const int SEG = TK_BINS / WAVE;             // 32 bins per lane
+    int h[SEG];
+    int seg_sum = 0;
+    for (int j = 0; j < SEG; ++j) {
+        h[j] = hist[lane * SEG + j];
+        seg_sum += h[j];
+    }
+    // suffix over lanes: sum of seg_sum for lanes strictly greater
+    int suffix_excl = 0;
+    {
+        int acc = seg_sum;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int other = __shfl_down(acc, off, WAVE);
+            if (lane + off < WAVE) acc += other;
+        }
+        // acc now = suffix inclusive at this lane; recover exclusive
+        suffix_excl = acc - seg_sum;
+    }
+    // walk own segment from the right: first j (largest) with suffix >= target
+    int found_b = -1, found_above = 0;
+    int acc = 0;
+    for (int j = SEG - 1; j >= 0; --j) {
+        int suffix_here = suffix_excl + acc + h[j];
+        if (suffix_here >= target) {
+            found_b = lane * SEG + j;
+            found_above = suffix_excl + acc;  // suffix(b+1)
+            break;
+        }
+        acc += h[j];
+    }
+    // max-reduce found_b across the wave; winning lane writes results
+    int best = found_b;
+    for (int off = 1; off < WAVE; off <<= 1) {
+        int other = __shfl_down(best, off, WAVE);
+        if (lane + off < WAVE && other > best) best = other;
+    }
+    best = __shfl(best, 0, WAVE);
+    if (found_b == best && found_b >= 0) {
         if (level == 1) {
             sc[0] = found_b;
             sc[1] = found_above;
@@ -944,9 +937,40 @@ __global__ void bt_thresh_kernel(const int* __restrict__ hist,
     const int* H = hist + (int64_t)t * TK_BINS;
     int* S = sc + t * 4;
     const int target = (level == 1) ? (int)k : (int)k - S[1];
-    int found_b, found_above;
-    tk_thresh_wave(H, target, &found_b, &found_above);
-    if (found_b >= 0) {  // winner lane
+    const int SEG = TK_BINS / WAVE;
+    int h[SEG];
+    int seg_sum = 0;
+    for (int j = 0; j < SEG; ++j) {
+        h[j] = H[lane * SEG + j];
+        seg_sum += h[j];
+    }
+    int suffix_excl;
+    {
+        int acc = seg_sum;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int other = __shfl_down(acc, off, WAVE);
+            if (lane + off < WAVE) acc += other;
+        }
+        suffix_excl = acc - seg_sum;
+    }
+    int found_b = -1, found_above = 0;
+    int acc = 0;
+    for (int j = SEG - 1; j >= 0; --j) {
+        int suffix_here = suffix_excl + acc + h[j];
+        if (suffix_here >= target) {
+            found_b = lane * SEG + j;
+            found_above = suffix_excl + acc;
+            break;
+        }
+        acc += h[j];
+    }
+    int best = found_b;
+    for (int off = 1; off < WAVE; off <<= 1) {
+        int other = __shfl_down(best, off, WAVE);
+        if (lane + off < WAVE && other > best) best = other;
+    }
+    best = __shfl(best, 0, WAVE);
+    if (found_b == best && found_b >= 0) {
         if (level == 1) {
             S[0] = found_b;
             S[1] = found_above;
@@ -1475,23 +1499,24 @@ torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
 
 #define PF_MAXD1 8  // degree <= 7
 
-// Shared per-segment polynomial fit body (one block): accumulate the
-// normalized-x power sums and moments, reduce through wave+LDS, build the
-// ridged Gram and solve in-register Cholesky; thread 0 writes the d1
-// coefficients.  Used by the per-tensor and batched fit kernels.
-#define PF_SH_COLS (2 * PF_MAXD1 - 1 + PF_MAXD1)
-__device__ void pf_fit_segment(const float* __restrict__ y, int64_t len, int degree,
-                               double (*sh)[PF_SH_COLS] /*[QBLOCK/WAVE]*/,
-                               double* __restrict__ out /*[d1]*/) {
+__global__ void polyfit_fit_kernel(const float* __restrict__ y,
+                                   const int64_t* __restrict__ seg_starts /*[S+1]*/,
+                                   int degree, double* __restrict__ coeffs /*[S,d1]*/) {
+    const int s = blockIdx.x;
+    const int64_t start = seg_starts[s];
+    const int64_t end = seg_starts[s + 1];
+    const int64_t len = end - start;
     const int d1 = degree + 1;
     const int np = 2 * degree + 1;
     const double inv_len = len > 0 ? 1.0 / (double)len : 1.0;
+
     double ps[2 * PF_MAXD1 - 1];
     double mo[PF_MAXD1];
     for (int p = 0; p < np; ++p) ps[p] = 0.0;
     for (int p = 0; p < d1; ++p) mo[p] = 0.0;
-    for (int64_t i = threadIdx.x; i < len; i += blockDim.x) {
-        const double x = (double)(i + 1) * inv_len;
+
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+        const double x = (double)(i - start + 1) * inv_len;
         const double yv = (double)y[i];
         double xp = 1.0;
         for (int p = 0; p < np; ++p) {
@@ -1500,11 +1525,13 @@ __device__ void pf_fit_segment(const float* __restrict__ y, int64_t len, int deg
             xp *= x;
         }
     }
+    // wave reduce, then cross-wave through LDS
     const int lane = threadIdx.x % WAVE, wid = threadIdx.x / WAVE;
     for (int off = WAVE / 2; off > 0; off >>= 1) {
         for (int p = 0; p < np; ++p) ps[p] += __shfl_down(ps[p], off, WAVE);
         for (int p = 0; p < d1; ++p) mo[p] += __shfl_down(mo[p], off, WAVE);
     }
+    __shared__ double sh[QBLOCK / WAVE][2 * PF_MAXD1 - 1 + PF_MAXD1];
     if (lane == 0) {
         for (int p = 0; p < np; ++p) sh[wid][p] = ps[p];
         for (int p = 0; p < d1; ++p) sh[wid][np + p] = mo[p];
@@ -1515,6 +1542,7 @@ __device__ void pf_fit_segment(const float* __restrict__ y, int64_t len, int deg
         for (int p = 0; p < np; ++p) ps[p] += sh[w][p];
         for (int p = 0; p < d1; ++p) mo[p] += sh[w][np + p];
     }
+    // gram[i][j] = S_{i+j}; ridge = max|diag| * 1e-10 + 1e-30
     double G[PF_MAXD1][PF_MAXD1];
     double dmax = 0.0;
     for (int i = 0; i < d1; ++i) {
@@ -1525,13 +1553,16 @@ __device__ void pf_fit_segment(const float* __restrict__ y, int64_t len, int deg
     for (int i = 0; i < d1; ++i)
         for (int j = 0; j < d1; ++j)
             G[i][j] = ps[i + j] + (i == j ? ridge : 0.0);
+    // in-register Cholesky solve G c = mo
     double L[PF_MAXD1][PF_MAXD1];
     for (int i = 0; i < d1; ++i) {
         for (int j = 0; j <= i; ++j) {
             double sum = G[i][j];
             for (int p = 0; p < j; ++p) sum -= L[i][p] * L[j][p];
-            if (i == j) L[i][j] = sqrt(sum > 1e-300 ? sum : 1e-300);
-            else L[i][j] = sum / L[j][j];
+            if (i == j)
+                L[i][j] = sqrt(sum > 1e-300 ? sum : 1e-300);
+            else
+                L[i][j] = sum / L[j][j];
         }
     }
     double yv[PF_MAXD1];
@@ -1540,24 +1571,12 @@ __device__ void pf_fit_segment(const float* __restrict__ y, int64_t len, int deg
         for (int p = 0; p < i; ++p) sum -= L[i][p] * yv[p];
         yv[i] = sum / L[i][i];
     }
-    double cr[PF_MAXD1];
+    double* c = coeffs + (int64_t)s * d1;
     for (int i = d1 - 1; i >= 0; --i) {
         double sum = yv[i];
-        for (int p = i + 1; p < d1; ++p) sum -= L[p][i] * cr[p];
-        cr[i] = sum / L[i][i];
+        for (int p = i + 1; p < d1; ++p) sum -= L[p][i] * c[p];
+        c[i] = sum / L[i][i];
     }
-    for (int i = 0; i < d1; ++i) out[i] = cr[i];
-}
-
-__global__ void polyfit_fit_kernel(const float* __restrict__ y,
-                                   const int64_t* __restrict__ seg_starts /*[S+1]*/,
-                                   int degree, double* __restrict__ coeffs /*[S,d1]*/) {
-    const int sgi = blockIdx.x;
-    const int64_t start = seg_starts[sgi];
-    const int64_t len = seg_starts[sgi + 1] - start;
-    __shared__ double sh[QBLOCK / WAVE][PF_SH_COLS];
-    pf_fit_segment(y + start, len, degree, sh,
-                   coeffs + (int64_t)sgi * (degree + 1));
 }
 
 // Horner evaluation of the fitted piecewise polynomial (binary search for
@@ -1790,13 +1809,79 @@ __global__ void bt2_fit_kernel(const float* __restrict__ sorted /*[T,kmax]*/,
                                int64_t kmax, uint8_t* __restrict__ wire) {
     const int t = seg_t[blockIdx.x];
     const int si = seg_i[blockIdx.x];
-    const int64_t* D = bt_row(desc, t);
+    const int64_t* D = bt2_row(desc, t);
     const int64_t* st = starts + (int64_t)t * PF_SMAX;
     const int64_t start = st[si];
-    const int64_t len = st[si + 1] - start;
-    double* c = (double*)(wire + D[12]) + (int64_t)si * (degree + 1);
-    __shared__ double sh[QBLOCK / WAVE][PF_SH_COLS];
-    pf_fit_segment(sorted + (int64_t)t * kmax + start, len, degree, sh, c);
+    const int64_t end = st[si + 1];
+    const int64_t len = end - start;
+    const int d1 = degree + 1;
+    const int np = 2 * degree + 1;
+    const double inv_len = len > 0 ? 1.0 / (double)len : 1.0;
+    const float* __restrict__ y = sorted + (int64_t)t * kmax;
+
+    double ps[2 * PF_MAXD1 - 1];
+    double mo[PF_MAXD1];
+    for (int p = 0; p < np; ++p) ps[p] = 0.0;
+    for (int p = 0; p < d1; ++p) mo[p] = 0.0;
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+        const double x = (double)(i - start + 1) * inv_len;
+        const double yv = (double)y[i];
+        double xp = 1.0;
+        for (int p = 0; p < np; ++p) {
+            ps[p] += xp;
+            if (p < d1) mo[p] += xp * yv;
+            xp *= x;
+        }
+    }
+    const int lane = threadIdx.x % WAVE, wid = threadIdx.x / WAVE;
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        for (int p = 0; p < np; ++p) ps[p] += __shfl_down(ps[p], off, WAVE);
+        for (int p = 0; p < d1; ++p) mo[p] += __shfl_down(mo[p], off, WAVE);
+    }
+    __shared__ double sh[QBLOCK / WAVE][2 * PF_MAXD1 - 1 + PF_MAXD1];
+    if (lane == 0) {
+        for (int p = 0; p < np; ++p) sh[wid][p] = ps[p];
+        for (int p = 0; p < d1; ++p) sh[wid][np + p] = mo[p];
+    }
+    __syncthreads();
+    if (threadIdx.x != 0) return;
+    for (int w = 1; w < (int)(blockDim.x / WAVE); ++w) {
+        for (int p = 0; p < np; ++p) ps[p] += sh[w][p];
+        for (int p = 0; p < d1; ++p) mo[p] += sh[w][np + p];
+    }
+    double G[PF_MAXD1][PF_MAXD1];
+    double dmax = 0.0;
+    for (int i = 0; i < d1; ++i) {
+        double dg = fabs(ps[2 * i]);
+        if (dg > dmax) dmax = dg;
+    }
+    const double ridge = dmax * 1e-10 + 1e-30;
+    for (int i = 0; i < d1; ++i)
+        for (int j = 0; j < d1; ++j)
+            G[i][j] = ps[i + j] + (i == j ? ridge : 0.0);
+    double L[PF_MAXD1][PF_MAXD1];
+    for (int i = 0; i < d1; ++i) {
+        for (int j = 0; j <= i; ++j) {
+            double sum = G[i][j];
+            for (int p = 0; p < j; ++p) sum -= L[i][p] * L[j][p];
+            if (i == j) L[i][j] = sqrt(sum > 1e-300 ? sum : 1e-300);
+            else L[i][j] = sum / L[j][j];
+        }
+    }
+    double yv[PF_MAXD1];
+    for (int i = 0; i < d1; ++i) {
+        double sum = mo[i];
+        for (int p = 0; p < i; ++p) sum -= L[i][p] * yv[p];
+        yv[i] = sum / L[i][i];
+    }
+    double* c = (double*)(wire + D[12]) + (int64_t)si * d1;
+    double cr[PF_MAXD1];
+    for (int i = d1 - 1; i >= 0; --i) {
+        double sum = yv[i];
+        for (int p = i + 1; p < d1; ++p) sum -= L[p][i] * cr[p];
+        cr[i] = sum / L[i][i];
+    }
+    for (int i = 0; i < d1; ++i) c[i] = cr[i];
 }
 
 // pack each tensor's mapping (argsort of the padded sort) at nbits into the

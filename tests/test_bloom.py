@@ -182,3 +182,54 @@ def test_fp16_wire_values_roundtrip():
     assert torch.equal(own[1], i2)
     assert torch.equal(own[0], v2)  # cache == wire decode (residual exact)
     assert torch.allclose(v2, t[i2], rtol=1e-3, atol=1e-4)
+
+
+def test_conflict_sets_deterministic_across_ranks():
+    """conflict_sets policy must re-derive the identical selection on every
+    rank from the wire alone (policies.hpp:136-146 determinism contract)."""
+    import torch
+
+    from deepreduce_amd.codecs import compressor
+
+    torch.manual_seed(4)
+    t = torch.randn(20_000)
+    k = 200
+    _, idxs = t.abs().topk(k)
+    params = {"policy": "conflict_sets", "policy_seed": 123, "dense_tensor": t}
+    v, bits, shape = compressor["bloom"].compress((t[idxs], idxs, t.size()), params)
+    params.pop("dense_tensor")
+    params.pop("_own_decoded", None)
+    outs = [compressor["bloom"].decompress((v.clone(), bits.clone(), shape),
+                                           dict(params)) for _ in range(3)]
+    for _, i2, _ in outs[1:]:
+        assert torch.equal(i2, outs[0][1])
+    assert outs[0][1].numel() == k
+
+
+def test_measured_fpr_ground_truth():
+    """measured_fpr == brute-force false-positive rate of the filter."""
+    import torch
+
+    from deepreduce_amd.codecs.bloom import Bloom
+    from deepreduce_amd.metrics import measured_fpr, policy_errors
+    from deepreduce_amd.ops import bloom_query_members
+
+    torch.manual_seed(5)
+    universe = 30_000
+    idxs = torch.randperm(universe)[:300]
+    params = {"policy": "leftmost", "fpr": 0.01}
+    v, bits, shape = Bloom.compress((torch.randn(300), idxs,
+                                     torch.Size([universe])), params)
+    from deepreduce_amd.codecs.bloom import get_bf_config
+
+    num_hash, m = get_bf_config(300, 0.01)
+    got = measured_fpr(bits, m, num_hash, universe, idxs)
+    members = bloom_query_members(bits, m, num_hash, torch.arange(universe))
+    true = torch.zeros(universe, dtype=torch.bool)
+    true[idxs] = True
+    brute = float((members & ~true).sum()) / float((~true).sum())
+    assert abs(got - brute) < 1e-9
+    # policy errors: leftmost selection vs true set
+    _, rec, _ = Bloom.decompress((v, bits, shape), params)
+    errs = policy_errors(rec, idxs)
+    assert errs == len(set(rec.tolist()) - set(idxs.tolist()))

@@ -365,3 +365,42 @@ def test_bench_driver_contract_2proc_cpu():
     assert d["n_gpus"] == 2 and d["metric"] == "images/sec"
     assert d["config"]["parallelism"] == "dp2"
     assert 0 < d["config"]["rel_volume"] < 0.2
+
+
+def _run_threshold_ragged(rank, world, q):
+    try:
+        _init(rank, world)
+        from deepreduce_amd import deepreduce_from_params
+
+        grc = deepreduce_from_params({
+            "compressor": "threshold", "threshold": 0.8,
+            "memory": "residual", "communicator": "allgather",
+        })
+        torch.manual_seed(800 + rank)  # different sparsity per rank -> ragged
+        g = torch.randn(6000)
+        out = grc.step_many([("w", g.clone())])[0]
+        outs = [torch.empty_like(out) for _ in range(world)]
+        dist.all_gather(outs, out)
+        ok = all(torch.allclose(outs[0], o, atol=1e-6) for o in outs)
+        q.put((rank, bool(ok)))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"ERROR: {e!r}"))
+
+
+def test_threshold_ragged_allgather():
+    """Per-rank payload sizes differ (threshold sparsifier): the two-phase
+    ragged exchange must still produce identical averages everywhere."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    os.environ["MASTER_PORT"] = "29781"
+    procs = [ctx.Process(target=_run_threshold_ragged, args=(r, world, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, ok in results:
+        assert ok is True, f"rank {rank}: {ok}"

@@ -293,3 +293,42 @@ def test_log_stats_wiring(tmp_path):
     assert len(recs) == 4  # 2 tensors x 2 steps
     assert {r["tensor"] for r in recs} == {"w", "b"}
     assert all(r["wire_bytes"] > 0 and r["dense_bytes"] > 0 for r in recs)
+
+
+def test_randomk_same_positions_across_ranks():
+    """RandomK must pick identical positions for the same (name, step) on
+    every rank (seed = hash(name)+step, tensorflow/deepreduce.py:290-298),
+    or decompressed averages diverge."""
+    from deepreduce_amd.compressors import RandomKCompressor
+
+    a, b = RandomKCompressor(0.05), RandomKCompressor(0.05)
+    t1 = torch.randn(4000)
+    t2 = torch.randn(4000)  # different values, same positions expected
+    (_, ia), _ = a.compress(t1, "layer.weight")
+    (_, ib), _ = b.compress(t2, "layer.weight")
+    assert torch.equal(ia, ib)
+    # next step: different positions than step 0, still rank-consistent
+    (_, ia2), _ = a.compress(t1, "layer.weight")
+    (_, ib2), _ = b.compress(t2, "layer.weight")
+    assert torch.equal(ia2, ib2)
+    assert not torch.equal(ia, ia2)
+
+
+def test_codec_edge_cases_tiny_k():
+    """k=1 and k=2 survive every registered codec round trip."""
+    from deepreduce_amd.codecs import compressor
+
+    shape = torch.Size([5000])
+    for name in ["bloom", "rle", "huffman", "pfor", "qsgd", "gzip"]:
+        for k in (1, 2):
+            vals = torch.randn(k)
+            idxs = torch.tensor(list(range(17, 17 + k)), dtype=torch.int64)
+            params = {"policy": "p0"} if name == "bloom" else {}
+            v, w, s = compressor[name].compress((vals.clone(), idxs.clone(), shape), params)
+            v2, i2, _ = compressor[name].decompress((v, w, s), params)
+            if name == "bloom":
+                assert set(idxs.tolist()) <= set(i2.tolist()), name
+            elif name in ("gzip", "qsgd"):
+                assert torch.equal(i2, idxs), name
+            else:
+                assert set(i2.tolist()) == set(idxs.tolist()), (name, k)

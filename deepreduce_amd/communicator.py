@@ -199,12 +199,13 @@ class Allgather(Communicator):
                 large = [(i, n, t) for i, (n, t) in enumerate(named_tensors)
                          if t.numel() > 1000]
                 out = [None] * len(named_tensors)
+                large_bytes = 0
                 if large:
                     for (i, _, _), r in zip(
                         large, self._step_many_impl([(n, t) for _, n, t in large])
                     ):
                         out[i] = r
-                large_bytes = self.last_wire_bytes
+                    large_bytes = self.last_wire_bytes
                 flat = torch.cat([t.reshape(-1) for _, _, t in small])
                 self.last_wire_bytes = large_bytes + flat.numel() * flat.element_size()
                 if self.world_size > 1:

@@ -188,6 +188,10 @@ def _run_fused_equivalence(rank, world, mode, q):
             params.update({"deepreduce": "both", "index": "bloom", "value": "polyfit"})
         elif mode == "value":
             params.update({"deepreduce": "value", "value": "polyfit"})
+        elif mode == "bothq":
+            params.update({"deepreduce": "both", "index": "bloom", "value": "qsgd"})
+        elif mode == "dexp":
+            params.update({"deepreduce": "value", "value": "doubleexp"})
         elif mode == "dense":
             params = {"compressor": "none", "memory": "none", "communicator": "allreduce"}
         grc_a = deepreduce_from_params(dict(params))
@@ -210,7 +214,8 @@ def _run_fused_equivalence(rank, world, mode, q):
         q.put((rank, f"ERROR: {e!r}"))
 
 
-@pytest.mark.parametrize("mode", ["plain", "index", "fp16", "both", "value", "dense"])
+@pytest.mark.parametrize("mode", ["plain", "index", "fp16", "both", "value",
+                                  "bothq", "dexp", "dense"])
 def test_fused_step_many_matches_per_tensor(mode):
     world = 2
     ctx = mp.get_context("spawn")

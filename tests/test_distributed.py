@@ -203,7 +203,10 @@ def _run_fused_equivalence(rank, world, mode, q):
         ok = True
         for step in range(3):  # multiple steps: residuals must evolve identically
             grads = [(n, (t * (step + 1)).clone()) for n, t in named]
+            # qsgd rounds stochastically: align the RNG draws of both paths
+            torch.manual_seed(9000 + step)
             fused = grc_a.step_many([(n, t.clone()) for n, t in grads])
+            torch.manual_seed(9000 + step)
             loop = [grc_b.step(t.clone(), n) for n, t in grads]
             for f, l in zip(fused, loop):
                 if not torch.allclose(f.reshape(-1), l.reshape(-1), atol=1e-5):

@@ -332,3 +332,32 @@ def test_codec_edge_cases_tiny_k():
                 assert torch.equal(i2, idxs), name
             else:
                 assert set(i2.tolist()) == set(idxs.tolist()), (name, k)
+
+
+def test_frozen_params_are_skipped():
+    """Models with frozen (requires_grad=False) or grad-less parameters
+    exchange only the live gradients."""
+    import torch.nn as nn
+
+    from deepreduce_amd import DistributedOptimizer, deepreduce_from_params
+
+    torch.manual_seed(1)
+    model = nn.Sequential(nn.Linear(2000, 50), nn.ReLU(), nn.Linear(50, 4))
+    model[0].weight.requires_grad_(False)
+    grc = deepreduce_from_params({
+        "compressor": "topk", "memory": "residual",
+        "communicator": "allgather", "compress_ratio": 0.05,
+        "deepreduce": "index", "index": "bloom", "policy": "leftmost",
+    })
+    opt = DistributedOptimizer(
+        torch.optim.SGD([p for p in model.parameters() if p.requires_grad],
+                        lr=0.1), grc, model)
+    frozen_before = model[0].weight.detach().clone()
+    for s in range(3):
+        x = torch.randn(8, 2000)
+        y = torch.randint(0, 4, (8,))
+        opt.zero_grad(set_to_none=False)
+        torch.nn.functional.cross_entropy(model(x), y).backward()
+        opt.step()
+    assert torch.equal(model[0].weight, frozen_before)
+    assert model[2].weight.grad is not None

@@ -361,3 +361,32 @@ def test_frozen_params_are_skipped():
         opt.step()
     assert torch.equal(model[0].weight, frozen_before)
     assert model[2].weight.grad is not None
+
+
+def test_bench_json_contract_single_process():
+    """bench.py's rank-0 JSON must carry every key the round driver
+    parses, with sane values (single-process CPU invocation)."""
+    import json
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--model", "resnet20", "--steps", "2",
+         "--warmup", "1", "--batch", "4", "--device", "cpu"],
+        capture_output=True, text=True, timeout=600,
+        cwd=__import__("os").path.dirname(__import__("os").path.dirname(
+            __import__("os").path.abspath(__file__))),
+    )
+    assert out.returncode == 0, out.stderr[-1500:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1  # exactly ONE JSON line
+    d = json.loads(lines[0])
+    for key in ["metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"]:
+        assert key in d, key
+    assert d["higher_is_better"] is True and d["scaling"] == "weak"
+    assert d["data"] == "synthetic" and d["n_gpus"] == 1
+    for key in ["model", "global_batch", "seq_len", "parallelism"]:
+        assert key in d["config"], key
+    assert d["value"] > 0 and d["ms_per_step"] > 0

@@ -390,3 +390,34 @@ def test_bench_json_contract_single_process():
     for key in ["model", "global_batch", "seq_len", "parallelism"]:
         assert key in d["config"], key
     assert d["value"] > 0 and d["ms_per_step"] > 0
+
+
+def test_broadcast_communicator_and_logger_csv(tmp_path):
+    """Broadcast communicator round trip (world=1) and the Logger-op
+    equivalent CSV dump (logger.cc parity)."""
+    from deepreduce_amd import deepreduce_from_params
+    from deepreduce_amd.metrics import StatsLogger
+
+    grc = deepreduce_from_params({
+        "compressor": "none", "memory": "none", "communicator": "broadcast"})
+    t = torch.randn(128)
+    out = grc.step(t.clone(), "w")
+    assert torch.equal(out, t)
+
+    st = StatsLogger(str(tmp_path))
+    fn = st.dump_values("conv1/weight", torch.tensor([1.5, -2.25]),
+                        torch.tensor([0.1, 0.2, 0.3]))
+    lines = open(fn).read().strip().splitlines()
+    assert lines[0] == "1.5,-2.25"
+    assert lines[1].startswith("0.1")
+
+
+def test_intpack_wide_values():
+    """n-bit pack/unpack at widths up to 63 bits."""
+    from deepreduce_amd.codecs.intpack import pack_with_header, unpack_with_header
+
+    for nbits in (1, 7, 24, 33, 48, 63):
+        v = torch.randint(0, 2, (257,), dtype=torch.int64) * ((1 << (nbits - 1)) - 1)
+        wire = pack_with_header(v, nbits=nbits)
+        out = unpack_with_header(wire)
+        assert torch.equal(out, v), nbits

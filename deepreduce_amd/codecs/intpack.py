@@ -22,6 +22,11 @@ def pack_with_header(values: torch.Tensor, nbits: int | None = None) -> torch.Te
     if nbits is None:
         mx = int(v.max().item()) if n else 0
         nbits = max(1, mx.bit_length())
+    if nbits > 57:
+        # a value may straddle 9 bytes beyond this width, which the
+        # 8-byte-window unpack kernels do not support; real wires
+        # (indices/gaps/runs/mappings) never exceed 31 bits
+        raise ValueError(f"nbits={nbits} unsupported (max 57)")
     stream = ops.pack_ints(v, nbits)
     header = torch.tensor(
         [n & 255, (n >> 8) & 255, (n >> 16) & 255, (n >> 24) & 255, nbits],

@@ -416,8 +416,12 @@ def test_intpack_wide_values():
     """n-bit pack/unpack at widths up to 63 bits."""
     from deepreduce_amd.codecs.intpack import pack_with_header, unpack_with_header
 
-    for nbits in (1, 7, 24, 33, 48, 63):
+    for nbits in (1, 7, 24, 33, 48, 57):
         v = torch.randint(0, 2, (257,), dtype=torch.int64) * ((1 << (nbits - 1)) - 1)
         wire = pack_with_header(v, nbits=nbits)
         out = unpack_with_header(wire)
         assert torch.equal(out, v), nbits
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError, match="max 57"):
+        pack_with_header(torch.tensor([1]), nbits=63)

@@ -29,4 +29,5 @@ from .factory import deepreduce_from_params, grace_from_params  # noqa: F401
 from .helper import tensor_bits  # noqa: F401
 from .memory import NoneMemory, ResidualMemory  # noqa: F401
 from .optimizer import DistributedOptimizer, reduce_gradients  # noqa: F401
+from .params import validate as validate_params  # noqa: F401
 from .wrappers import DeepReduce, IndexCompressor, ValueCompressor, deepreduce_wrapper  # noqa: F401

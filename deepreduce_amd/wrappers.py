@@ -168,6 +168,7 @@ class IndexCompressor(_WrapperBase):
             or self.params.get("policy", "leftmost") != "leftmost"
             or numel <= _BYPASS_NUMEL
             or len(payloads) < 2
+            or len(payloads) > 16  # MAXR of the batched query kernel
         ):
             return None
         from . import ops

@@ -117,9 +117,9 @@ class Communicator:
         comp = self.compressor
         if t.numel() <= 1000 and getattr(self, "params", {}).get("small_dense", True):
             return t.numel() * 4
-        metas = getattr(self, "_bt_pipeline", None)
-        if metas is not None:
-            bp = metas[1]
+        cached = getattr(self, "_bt_pipeline", None)
+        if cached is not None:
+            bp = cached[1]
             if name in bp.names:
                 i = bp.names.index(name)
                 return sum(

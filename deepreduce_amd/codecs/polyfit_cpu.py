@@ -19,24 +19,38 @@ from . import SparseCompressor
 
 
 def find_breaks(curve: np.ndarray, num_of_breaks: int = 10):
-    """Recursive max-chord-distance knot selection (ascending input).
+    """Greedy chord-split knot selection over successive suffixes.
 
-    Parity with pytorch/deepreduce.py:566-582.
+    Behavioral contract (same algorithm as the reference layer,
+    pytorch/deepreduce.py:566-582 — paper Lemma 1 — expressed in this
+    codebase's idiom): walk the sorted curve left to right.  At each step,
+    draw the straight segment joining the current window's first and last
+    points (NOTE: the reference anchors the chord's right end at the
+    *global* last sample via y[-1] of the suffix — the suffix always ends
+    at the curve's end, so these coincide), place a knot where the curve
+    deviates most from that segment, then continue on the tail to the
+    right of the knot.  A window shorter than 20*num_of_breaks samples
+    never receives a knot (min-segment guard), and the walk also stops
+    once the remaining tail falls under the same guard.  Knots are
+    absolute indices into `curve`, strictly increasing.
     """
-    y = curve
-    breaks = []
-    break_index = 0
+    n = len(curve)
+    guard = 20 * num_of_breaks
+    knots: list[int] = []
+    start = 0
     for _ in range(num_of_breaks):
-        if len(y) < 20 * num_of_breaks:
+        window = n - start
+        if window < guard:
             break
-        line = np.linspace(y[0], y[-1], len(y))
-        distance = np.abs(line - y)
-        break_index += int(np.argmax(distance))
-        if (len(curve) - break_index) < 20 * num_of_breaks:
+        left, right = curve[start], curve[n - 1]
+        # chord sampled at the window's integer positions
+        chord = left + (right - left) * (np.arange(window) / max(window - 1, 1))
+        k = start + int(np.argmax(np.abs(chord - curve[start:])))
+        if n - k < guard:
             break
-        breaks.append(break_index)
-        y = curve[break_index:]
-    return breaks
+        knots.append(k)
+        start = k
+    return knots
 
 
 def _fit(curve: np.ndarray, breaks, degree: int):

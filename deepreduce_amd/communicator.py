@@ -419,11 +419,32 @@ class Allreduce(Communicator):
 
     def step_many(self, named_tensors):
         """Dense fused path: ONE flat all-reduce for the whole model (the
-        classic flat-bucket DDP exchange — the RCCL baseline bench)."""
+        classic flat-bucket DDP exchange — the RCCL baseline bench).
+
+        Residual semantics match the per-tensor `step`: the residual is
+        updated against the rank's OWN decode (compensated − D(C(t))), not
+        the globally averaged result — for the 'none' compressor that makes
+        the residual exactly zero, and for a lossy compressor it is the
+        standard error-feedback update.  What travels is each rank's dense
+        own-decode, exactly as `send_receive` ships `decompress(tensors)`.
+        """
+        from .compressors import NoneCompressor
+
         names = [n for n, _ in named_tensors]
         grads = [t for _, t in named_tensors]
+        comp = self.compressor
         compensated = self.memory.compensate_many(grads, names)
-        flat = torch.cat([t.reshape(-1) for t in compensated])
+        if isinstance(comp, NoneCompressor):
+            decs = compensated  # D(C(t)) = t: residual -> 0
+        else:
+            own_decode = getattr(comp, "decompress_own", None)
+            decs = []
+            for n, t in zip(names, compensated):
+                tc, ctx = comp.compress(t, n)
+                d = own_decode(tc, ctx, n) if own_decode else comp.decompress(tc, ctx)
+                decs.append(d.view_as(t))
+        self.memory.update_many(compensated, names, decs)
+        flat = torch.cat([t.reshape(-1) for t in decs])
         world = self.world_size
         self.last_wire_bytes = flat.numel() * flat.element_size()
         if world > 1:
@@ -436,9 +457,6 @@ class Allreduce(Communicator):
             n = g.numel()
             outs.append(flat[off : off + n].view(g.shape))
             off += n
-        # residual memory is a no-op for the dense baseline ('none'), but
-        # honor it if configured
-        self.memory.update_many(compensated, names, outs)
         return outs
 
 

@@ -94,6 +94,8 @@ def _graph_safe(grc) -> bool:
         return False
     if params.get("micro-benchmark"):
         return False  # timing prints sync
+    if params.get("log_stats"):
+        return False  # host-side JSONL logger would freeze inside the graph
     return True
 
 
@@ -155,12 +157,20 @@ class DistributedOptimizer:
         # set_to_none=False keeps them), residual buffers (ResidualMemory
         # updates in place), and graph-pool intermediates.  Canonical
         # recipe: warm the exact capture path up on a side stream first.
+        # The warmup run mutates live state (p.grad <- reduced values,
+        # residual updated), so snapshot grads + residuals before it and
+        # restore before capturing: the captured pass then compresses the
+        # pristine gradients exactly once, and the residual sees exactly
+        # one update this step.
         try:
+            grads, gsnap, rsnap = self._snapshot_state()
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
                 reduce_gradients(self.model, self.grc)
             torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            self._restore_state(grads, gsnap, rsnap)
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
@@ -173,6 +183,33 @@ class DistributedOptimizer:
             self._graph = None
             torch.cuda.synchronize()
             self.last_wire_bytes = reduce_gradients(self.model, self.grc)
+
+    def _snapshot_state(self):
+        """Clone p.grad buffers and residual-memory tensors (pre-warmup)."""
+        grads = [p.grad for p in self.model.parameters() if p.grad is not None]
+        gsnap = [g.detach().clone() for g in grads]
+        mem = getattr(self.grc, "memory", None)
+        rsnap = None
+        if mem is not None and hasattr(mem, "residuals"):
+            rsnap = {k: v.detach().clone() for k, v in mem.residuals.items()}
+        return grads, gsnap, rsnap
+
+    def _restore_state(self, grads, gsnap, rsnap):
+        for g, s in zip(grads, gsnap):
+            g.copy_(s)
+        mem = getattr(self.grc, "memory", None)
+        if rsnap is not None and mem is not None:
+            for k, s in rsnap.items():
+                r = mem.residuals.get(k)
+                if r is not None and r.shape == s.shape:
+                    r.copy_(s)
+            # names whose residual buffers appeared during warmup (first
+            # flat-pool build) started the warmup at zero: reset them
+            for k, r in mem.residuals.items():
+                if k not in rsnap:
+                    r.zero_()
+            for pool in getattr(mem, "_pools", {}).values():
+                pool["c"] = None  # transient alias: never reuse post-restore
 
     def _grad_ptrs(self):
         return tuple(p.grad.data_ptr() for p in self.model.parameters()

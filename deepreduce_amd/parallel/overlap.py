@@ -114,10 +114,13 @@ class OverlappedReducer:
                 outs = impl(named) if impl else [grc.step(t, n) for n, t in named]
                 self._launched[b] = ("done", outs, grc.last_wire_bytes)
                 return
+            from ..memory import ResidualMemory
+
             names = [n for n, _ in named]
             grads = [t for _, t in named]
-            compensated = grc.memory.compensate_many(grads, names)
-            flat_c = getattr(grc.memory, "_flat_c", None)
+            mem = grc.memory
+            compensated = mem.compensate_many(grads, names)
+            flat_c = getattr(mem, "_flat_c", None)
             if (flat_c is not None and compensated
                     and compensated[0].data_ptr() == flat_c.data_ptr()
                     and flat_c.numel() == bp.total_values):
@@ -125,7 +128,21 @@ class OverlappedReducer:
             else:
                 c_flat = torch.cat([t.reshape(-1) for t in compensated])
             wire, own = bp.compress_and_own(c_flat)
-            torch.sub(c_flat, own, out=grc.memory._flat_r)
+            # residual <- compensated - own decode; same guards as
+            # communicator._step_many_batched: the in-place fast path is only
+            # valid when c_flat IS this bucket's flat pool (same storage, so
+            # _flat_r is the matching residual buffer); otherwise fall back to
+            # the generic per-name update.  NoneMemory: nothing to update.
+            if isinstance(mem, ResidualMemory):
+                if c_flat is flat_c:
+                    torch.sub(c_flat, own, out=mem._flat_r)
+                else:
+                    offs = 0
+                    decs = []
+                    for t in grads:
+                        decs.append(own[offs : offs + t.numel()].view(t.shape))
+                        offs += t.numel()
+                    mem.update_many(compensated, names, decs)
             world = grc.world_size
             if world == 1:
                 self._launched[b] = ("own", bp, own, wire.numel())
@@ -150,7 +167,23 @@ class OverlappedReducer:
             for b, bucket in enumerate(self._buckets):
                 entry = self._launched.pop(b, None)
                 if entry is None:
+                    # Bucket never filled this step.  A PARTIAL arrival count
+                    # (frozen layer / conditional branch produced only some of
+                    # the bucket's grads) must not leak into the next step, or
+                    # the bucket would launch mid-backward on a stale mix —
+                    # exchange whatever grads exist now, synchronously, and
+                    # reset the counter.
+                    if self._arrived[b]:
+                        self._arrived[b] = 0
+                        present = [(n, p) for n, p in bucket
+                                   if p.grad is not None]
+                        for n, p in present:
+                            r = grc.step(p.grad.data.float(), n)
+                            p.grad.data.copy_(r.view_as(p.grad.data))
+                            self.last_wire_bytes += getattr(
+                                grc, "last_wire_bytes", 0)
                     continue  # no grads this step
+                self._arrived[b] = 0  # defensive: full reset each step
                 if entry[0] == "done":
                     _, outs, wb = entry
                     self.last_wire_bytes += wb

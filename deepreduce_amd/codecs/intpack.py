@@ -43,12 +43,224 @@ def unpack_with_header(wire: torch.Tensor) -> torch.Tensor:
     return ops.unpack_ints(wire[5:], n, nbits)
 
 
-class PFor(SparseCompressor):
-    """Sorted-delta + bit-pack index codec ('pfor').
+# ---------------------------------------------------------------------------
+# Block-PFoR (patched frame-of-reference) — the FastPFor-family codec the
+# reference binds by name (/root/reference/tensorflow/integer_compression.cc:
+# 62,161: delta + bit-packing with patched exceptions, achieved-ratio print
+# at :71-73).  Redesigned here, not transcribed: 128-int blocks, per-block
+# bit width chosen to minimize total bytes, outliers patched as
+# (position byte, packed high bits).  All tensor math — runs resident on
+# GPU (no per-block python loop; one host sync at compress for allocation
+# sizing, which is why the codec is not in the hipGraph-safe set).
+#
+# Wire format (uint8):
+#   [0:4]  n        count, LE u32
+#   [4]    0xFF     format tag (legacy fixed-width wires carry nbits<=57 here)
+#   [5:9]  nb       number of 128-int blocks, LE u32
+#   nb * 3 bytes    per-block meta: [width b][n_exceptions][exception width]
+#   per block:      ceil(cnt*b/8) bytes   low-b-bit stream (cnt=128, last may
+#                                         be partial)
+#   per block:      n_exc bytes           exception lane positions (0..127)
+#                   ceil(n_exc*exc_w/8)   exception high bits (v >> b)
+# ---------------------------------------------------------------------------
 
-    compress: sort idxs ascending (vals permuted to match), deltas =
-    [idx0, idx1-idx0-1, ...] (gaps), bit-packed at the max gap width.
-    Order-preserving w.r.t. ascending index order only.
+_PFOR_BLOCK = 128
+_PFOR_TAG = 0xFF
+
+
+def _bitlen(v: torch.Tensor) -> torch.Tensor:
+    """Element-wise bit length (0 for 0), exact integer comparisons."""
+    thresholds = (1 << torch.arange(33, dtype=torch.int64, device=v.device))
+    return (v.unsqueeze(-1) >= thresholds).sum(-1)
+
+
+def _scatter_bits(bit_arr: torch.Tensor, values: torch.Tensor,
+                  start_bits: torch.Tensor, widths: torch.Tensor):
+    """Write `widths[i]` low bits of values[i] at absolute bit offsets
+    start_bits[i] (LSB-first) into the uint8 0/1 array `bit_arr`."""
+    max_w = int(widths.max().item()) if widths.numel() else 0
+    for j in range(max_w):
+        mask = widths > j
+        if not bool(mask.any()):
+            break
+        pos = start_bits[mask] + j
+        bit_arr[pos] = ((values[mask] >> j) & 1).to(torch.uint8)
+
+
+def _gather_bits(bit_arr: torch.Tensor, start_bits: torch.Tensor,
+                 widths: torch.Tensor) -> torch.Tensor:
+    """Inverse of _scatter_bits: read widths[i] bits from start_bits[i]."""
+    out = torch.zeros(start_bits.numel(), dtype=torch.int64,
+                      device=bit_arr.device)
+    max_w = int(widths.max().item()) if widths.numel() else 0
+    for j in range(max_w):
+        mask = widths > j
+        if not bool(mask.any()):
+            break
+        pos = start_bits[mask] + j
+        out[mask] |= bit_arr[pos].long() << j
+    return out
+
+
+def _bits_to_bytes(bit_arr: torch.Tensor) -> torch.Tensor:
+    w = (1 << torch.arange(8, dtype=torch.int64, device=bit_arr.device))
+    return (bit_arr.view(-1, 8).long() * w).sum(-1).to(torch.uint8)
+
+
+def _bytes_to_bits(bytes_t: torch.Tensor) -> torch.Tensor:
+    shifts = torch.arange(8, dtype=torch.uint8, device=bytes_t.device)
+    return ((bytes_t.unsqueeze(-1) >> shifts) & 1).reshape(-1)
+
+
+def pfor_encode(ints: torch.Tensor) -> torch.Tensor:
+    """Encode non-negative int64 tensor into the block-PFoR wire (uint8)."""
+    v = ints.long().reshape(-1)
+    n = v.numel()
+    dev = v.device
+    nb = (n + _PFOR_BLOCK - 1) // _PFOR_BLOCK
+    header = torch.tensor(
+        [n & 255, (n >> 8) & 255, (n >> 16) & 255, (n >> 24) & 255, _PFOR_TAG,
+         nb & 255, (nb >> 8) & 255, (nb >> 16) & 255, (nb >> 24) & 255],
+        dtype=torch.uint8, device=dev)
+    if n == 0:
+        return header
+
+    pad = nb * _PFOR_BLOCK - n
+    vp = torch.nn.functional.pad(v, (0, pad)).view(nb, _PFOR_BLOCK)
+    cnt = torch.full((nb,), _PFOR_BLOCK, dtype=torch.int64, device=dev)
+    if pad:
+        cnt[-1] = _PFOR_BLOCK - pad
+    bl = _bitlen(vp)                                   # [nb, 128]
+    if pad:  # padding lanes must never become exceptions
+        lane = torch.arange(_PFOR_BLOCK, device=dev)
+        bl[-1] = torch.where(lane < cnt[-1], bl[-1], torch.zeros_like(bl[-1]))
+    maxbl = bl.max(dim=1).values                       # [nb]
+
+    # per-block histogram of bit lengths -> exceptions(b) = #(bl > b)
+    hist = torch.zeros(nb, 34, dtype=torch.int64, device=dev)
+    hist.scatter_add_(1, bl, torch.ones_like(bl))
+    suffix = hist.flip(1).cumsum(1).flip(1)            # suffix[b] = #(bl >= b)
+    bs = torch.arange(33, dtype=torch.int64, device=dev)   # candidate widths
+    ne = suffix[:, 1:34]                               # ne[:, b] = #(bl > b)
+    exw = (maxbl.unsqueeze(1) - bs.unsqueeze(0)).clamp_min(0)  # [nb, 33]
+    stream_bytes = (cnt.unsqueeze(1) * bs.unsqueeze(0) + 7) // 8
+    exc_bytes = ne + (ne * exw + 7) // 8
+    cost = stream_bytes + exc_bytes                    # [nb, 33] bytes
+    b = cost.argmin(dim=1)                             # [nb] chosen width
+    ne_b = ne.gather(1, b.unsqueeze(1)).squeeze(1)
+    exw_b = torch.where(ne_b > 0,
+                        exw.gather(1, b.unsqueeze(1)).squeeze(1),
+                        torch.zeros_like(b))
+
+    # ---- layout (host sync: allocation sizes) ----
+    sb = ((cnt * b + 7) // 8)                          # low-bit stream bytes
+    pe = ne_b                                          # position bytes
+    hb = (ne_b * exw_b + 7) // 8                       # high-bit bytes
+    sb_off = torch.cumsum(sb, 0) - sb
+    total_sb = int(sb.sum().item())
+    exc_off = torch.cumsum(pe + hb, 0) - (pe + hb)
+    total_exc = int((pe + hb).sum().item())
+
+    # ---- low-bit stream ----
+    lane = torch.arange(_PFOR_BLOCK, device=dev).unsqueeze(0).expand(nb, -1)
+    valid = lane < cnt.unsqueeze(1)
+    start_bits = (sb_off.unsqueeze(1) * 8 + lane * b.unsqueeze(1))[valid]
+    widths = b.unsqueeze(1).expand(nb, _PFOR_BLOCK)[valid]
+    low_mask = (1 << b).sub(1).unsqueeze(1).expand(nb, _PFOR_BLOCK)[valid]
+    bit_arr = torch.zeros(total_sb * 8, dtype=torch.uint8, device=dev)
+    _scatter_bits(bit_arr, vp[valid] & low_mask, start_bits, widths)
+    stream = _bits_to_bytes(bit_arr)
+
+    # ---- exceptions ----
+    is_exc = valid & (bl > b.unsqueeze(1))
+    exc_buf = torch.zeros(total_exc, dtype=torch.uint8, device=dev)
+    if bool(is_exc.any()):
+        flat_exc = is_exc.reshape(-1)
+        rank = torch.cumsum(flat_exc.long(), 0).view(nb, _PFOR_BLOCK)
+        ne_before = torch.cumsum(ne_b, 0) - ne_b
+        within = (rank - 1 - ne_before.unsqueeze(1))[is_exc]
+        blk = torch.arange(nb, device=dev).unsqueeze(1).expand(nb, _PFOR_BLOCK)[is_exc]
+        # positions
+        exc_buf[exc_off[blk] + within] = lane[is_exc].to(torch.uint8)
+        # high bits: a per-block sub-bit-array, byte aligned after positions
+        hi_bit_arr = torch.zeros(int(hb.sum().item()) * 8, dtype=torch.uint8,
+                                 device=dev)
+        hb_off = torch.cumsum(hb, 0) - hb
+        hi_start = hb_off[blk] * 8 + within * exw_b[blk]
+        _scatter_bits(hi_bit_arr, vp[is_exc] >> b[blk], hi_start, exw_b[blk])
+        hi_bytes = _bits_to_bytes(hi_bit_arr)
+        # interleave: per block [positions][high bytes]
+        hpos = exc_off[blk] + within  # position byte slots already written
+        hb_slot_off = exc_off + pe    # high bytes start after positions
+        byte_blk = torch.repeat_interleave(torch.arange(nb, device=dev), hb)
+        byte_within = torch.arange(hi_bytes.numel(), device=dev) - \
+            torch.repeat_interleave(hb_off, hb)
+        exc_buf[hb_slot_off[byte_blk] + byte_within] = hi_bytes
+
+    meta = torch.stack([b, ne_b, exw_b], dim=1).reshape(-1).to(torch.uint8)
+    return torch.cat([header, meta, stream, exc_buf])
+
+
+def pfor_decode(wire: torch.Tensor) -> torch.Tensor:
+    dev = wire.device
+    head = wire[:9].cpu()
+    n = int(head[0]) | (int(head[1]) << 8) | (int(head[2]) << 16) | (int(head[3]) << 24)
+    assert int(head[4]) == _PFOR_TAG, "not a block-PFoR wire"
+    nb = int(head[5]) | (int(head[6]) << 8) | (int(head[7]) << 16) | (int(head[8]) << 24)
+    if n == 0:
+        return torch.zeros(0, dtype=torch.int64, device=dev)
+    meta = wire[9 : 9 + nb * 3].to(dev).long().view(nb, 3)
+    b, ne_b, exw_b = meta[:, 0], meta[:, 1], meta[:, 2]
+    cnt = torch.full((nb,), _PFOR_BLOCK, dtype=torch.int64, device=dev)
+    if n % _PFOR_BLOCK:
+        cnt[-1] = n % _PFOR_BLOCK
+    sb = (cnt * b + 7) // 8
+    pe = ne_b
+    hb = (ne_b * exw_b + 7) // 8
+    sb_off = torch.cumsum(sb, 0) - sb
+    total_sb = int(sb.sum().item())
+    stream = wire[9 + nb * 3 : 9 + nb * 3 + total_sb].to(dev)
+    exc_buf = wire[9 + nb * 3 + total_sb :].to(dev)
+
+    bit_arr = _bytes_to_bits(stream)
+    lane = torch.arange(_PFOR_BLOCK, device=dev).unsqueeze(0).expand(nb, -1)
+    valid = lane < cnt.unsqueeze(1)
+    start_bits = (sb_off.unsqueeze(1) * 8 + lane * b.unsqueeze(1))[valid]
+    widths = b.unsqueeze(1).expand(nb, _PFOR_BLOCK)[valid]
+    out = _gather_bits(bit_arr, start_bits, widths)
+
+    total_ne = int(ne_b.sum().item())
+    if total_ne:
+        exc_off = torch.cumsum(pe + hb, 0) - (pe + hb)
+        blk = torch.repeat_interleave(torch.arange(nb, device=dev), ne_b)
+        ne_before = torch.cumsum(ne_b, 0) - ne_b
+        within = torch.arange(total_ne, device=dev) - ne_before[blk]
+        pos = exc_buf[exc_off[blk] + within].long()
+        hb_off = torch.cumsum(hb, 0) - hb
+        # rebuild the per-block high-bit sub-array
+        hi_bytes = torch.zeros(int(hb.sum().item()), dtype=torch.uint8,
+                               device=dev)
+        byte_blk = torch.repeat_interleave(torch.arange(nb, device=dev), hb)
+        byte_within = torch.arange(hi_bytes.numel(), device=dev) - \
+            torch.repeat_interleave(hb_off, hb)
+        hi_bytes[byte_within + hb_off[byte_blk]] = \
+            exc_buf[(exc_off + pe)[byte_blk] + byte_within]
+        hi_bits = _bytes_to_bits(hi_bytes)
+        hi = _gather_bits(hi_bits, hb_off[blk] * 8 + within * exw_b[blk],
+                          exw_b[blk])
+        # only the LAST block can be partial, so the flat output offset of
+        # (block, lane) is simply block*128 + lane
+        out[blk * _PFOR_BLOCK + pos] |= hi << b[blk]
+    return out
+
+
+class PFor(SparseCompressor):
+    """Block-PFoR index codec ('pfor'): sort ascending, delta to gaps,
+    patched frame-of-reference encode (128-int blocks, per-block widths,
+    exception patching — see pfor_encode).  Order-preserving w.r.t.
+    ascending index order only.  Achieved bits/int is reported via
+    params['_pfor_bits_per_int'] and printed under 'micro-benchmark'
+    (reference ratio print: integer_compression.cc:71-73).
     """
 
     order_preserving = False
@@ -62,13 +274,19 @@ class PFor(SparseCompressor):
             gaps = torch.cat([s[:1], s[1:] - s[:-1] - 1])
         else:
             gaps = s
-        wire = pack_with_header(gaps)
+        wire = pfor_encode(gaps)
+        if s.numel():
+            bpi = wire.numel() * 8.0 / s.numel()
+            params["_pfor_bits_per_int"] = bpi
+            if params.get("micro-benchmark"):
+                print(f"pfor: {bpi:.2f} bits/int "
+                      f"({wire.numel()}B for {s.numel()} ints)")
         return vals, wire, shape
 
     @staticmethod
     def decompress(sparse_tensor, params):
         vals, wire, shape = sparse_tensor
-        gaps = unpack_with_header(wire).to(vals.device)
+        gaps = pfor_decode(wire).to(vals.device)
         if gaps.numel():
             idxs = torch.cumsum(gaps + 1, 0) - 1
         else:

@@ -67,4 +67,19 @@ def validate(params: dict) -> dict:
     fpr = params.get("fpr")
     if fpr is not None and not (0 < fpr < 1):
         raise ValueError(f"fpr must be in (0, 1), got {fpr}")
+    if params.get("policy") == "conflict_sets":
+        # The conflict-sets policy (reference policies.hpp:43-146 semantics)
+        # is a HOST-side sequential round-robin: inherently unvectorizable
+        # (each pick depends on every prior pick through the shared LCG and
+        # set-emptying), so it runs as python over the positive set with a
+        # device sync per tensor.  Correct and deterministic, but not a
+        # hot-path policy on MI355X — use 'leftmost' (sync-free fused
+        # kernel) or 'p0' for production runs.
+        import warnings
+
+        warnings.warn(
+            "policy='conflict_sets' is a debug/compatibility policy that "
+            "runs on the host (one device sync per tensor per step); use "
+            "'leftmost' or 'p0' on the hot path",
+            RuntimeWarning, stacklevel=3)
     return params

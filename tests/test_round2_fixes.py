@@ -259,3 +259,50 @@ def test_find_breaks_golden():
     assert 650 <= got[0] <= 900  # knot lands at/after the kink region
     # short curves: guarded, no knots
     assert find_breaks(np.arange(50, dtype=np.float64), 10) == []
+
+
+# ---- block-PFoR ------------------------------------------
+
+class TestBlockPFor:
+    def test_roundtrip_shapes_and_outliers(self):
+        from deepreduce_amd.codecs.intpack import pfor_decode, pfor_encode
+
+        torch.manual_seed(0)
+        for n in [0, 1, 127, 128, 129, 1000]:
+            v = torch.randint(0, 1 << 24, (n,), dtype=torch.int64)
+            assert torch.equal(pfor_decode(pfor_encode(v)), v)
+        # outlier patching: one huge value must not inflate the block width
+        g = torch.randint(0, 64, (4096,), dtype=torch.int64)
+        g[100] = 1 << 27
+        g[3000] = (1 << 31) - 1
+        w = pfor_encode(g)
+        assert torch.equal(pfor_decode(w), g)
+        bits_per_int = w.numel() * 8 / g.numel()
+        assert bits_per_int < 9, f"outliers not patched: {bits_per_int:.1f} b/int"
+
+    def test_beats_fixed_width_on_outlier_gaps(self):
+        from deepreduce_amd.codecs.intpack import pack_with_header, pfor_encode
+
+        torch.manual_seed(1)
+        g = torch.randint(0, 100, (5000,), dtype=torch.int64)
+        g[torch.randint(0, 5000, (40,))] = torch.randint(
+            1 << 20, 1 << 28, (40,), dtype=torch.int64)
+        assert pfor_encode(g).numel() * 3 < pack_with_header(g).numel(), \
+            "block-PFoR should be >=3x smaller than global fixed width here"
+
+    def test_codec_reports_bits_per_int(self):
+        from deepreduce_amd.codecs import compressor as registry
+
+        torch.manual_seed(2)
+        t = torch.randn(100000)
+        k = 1000
+        _, idx = t.abs().topk(k)
+        vals = t[idx]
+        params = {}
+        v, w, s = registry["pfor"].compress((vals, idx, t.size()), params)
+        assert 0 < params["_pfor_bits_per_int"] < 32
+        v2, i2, _ = registry["pfor"].decompress((v, w, t.size()), params)
+        assert torch.equal(i2.sort().values, idx.sort().values)
+        got = torch.zeros_like(t).scatter_(0, i2, v2)
+        want = torch.zeros_like(t).scatter_(0, idx, vals)
+        assert torch.allclose(got, want)

@@ -94,7 +94,10 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
     use_cuda = torch.cuda.is_available() and args.device != "cpu"
-    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    # modulo wrap: ranks beyond the device count co-locate (RCCL supports
+    # multiple ranks per GPU), so `--gpus 2` is testable on a 1-GPU box
+    ndev = max(1, torch.cuda.device_count()) if use_cuda else 1
+    device = torch.device(f"cuda:{local_rank % ndev}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
 

@@ -102,8 +102,18 @@ def main():
         torch.cuda.set_device(device)
 
     if env_world > 1:
-        backend = "nccl" if use_cuda else "gloo"
-        dist.init_process_group(backend=backend)
+        if use_cuda and env_world <= torch.cuda.device_count():
+            dist.init_process_group(backend="nccl")  # = RCCL on ROCm
+        elif use_cuda:
+            # more ranks than GPUs: RCCL refuses co-located ranks, so run
+            # the GPU pipeline with CPU-staged gloo transport (launch-
+            # qualification on small boxes; not a performance mode)
+            from deepreduce_amd.testing import stage_collectives_via_cpu
+
+            dist.init_process_group(backend="gloo")
+            stage_collectives_via_cpu()
+        else:
+            dist.init_process_group(backend="gloo")
     world = env_world
 
     from deepreduce_amd.models import registry

@@ -175,6 +175,10 @@ class DistributedOptimizer:
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
                 self._graph_wire_bytes = reduce_gradients(self.model, self.grc)
+            # capture RECORDS without executing: replay once so this step's
+            # exchange actually happens (on the restored, pristine grads —
+            # exactly one compress + one residual update this step)
+            g.replay()
             self._graph = g
             self._graph_grad_ptrs = self._grad_ptrs()
             self.last_wire_bytes = self._graph_wire_bytes

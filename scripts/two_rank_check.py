@@ -144,12 +144,22 @@ def run_config(name, params, device, rank, steps=3, overlap=False,
 
 def main():
     rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
     if torch.cuda.is_available():
         ndev = max(1, torch.cuda.device_count())
         device = torch.device(f"cuda:{rank % ndev}")
         torch.cuda.set_device(device)
-        dist.init_process_group("nccl")
-    else:  # logic dry-run on CPU (gloo) — the GPU test uses nccl
+        if world <= ndev:
+            dist.init_process_group("nccl")
+        else:
+            # RCCL refuses co-located ranks ("Duplicate GPU detected"):
+            # full GPU pipeline per rank, CPU-staged gloo transport
+            from deepreduce_amd.testing import stage_collectives_via_cpu
+
+            dist.init_process_group("gloo")
+            stage_collectives_via_cpu()
+            log(rank, "transport: CPU-staged gloo (ranks > GPUs)")
+    else:  # logic dry-run on CPU (gloo)
         device = torch.device("cpu")
         dist.init_process_group("gloo")
 

@@ -160,6 +160,16 @@ def cholesky_solve_small(G: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return torch.linalg.solve(G, b.unsqueeze(-1)).squeeze(-1)
 
 
+def dexp_fit(y: torch.Tensor, offs: torch.Tensor, lens: torch.Tensor) -> torch.Tensor:
+    """Fused DoubleExp cumulative-integral fit: y float32 (sorted ascending
+    per tensor), offs/lens int64 [B] -> coeffs float64 [B, 4] = (a, b, c, d).
+    GPU: one block per tensor (dexp_fit_kernel); CPU callers use the torch
+    fp64 reference in codecs/doubleexp.py directly."""
+    if _want_hip(y):
+        return _hip.dexp_fit(y, offs, lens)
+    raise RuntimeError("dexp_fit is a GPU op; use codecs.doubleexp torch path on CPU")
+
+
 def qsgd_quantize(vals: torch.Tensor, quantum_num: int, bucket_size: int):
     if _want_hip(vals):
         return _hip.qsgd_quantize(vals, quantum_num, bucket_size)

@@ -17,10 +17,22 @@ from __future__ import annotations
 import torch
 
 BT_CHUNK = 8192
+# dynamic-LDS budget for the compress-side bloom query (bt_qcount R=1):
+# 64 KB leaves 2 workgroups per CU (160 KB LDS) and covers the flagship's
+# 10-45 KB filters
+LDSQ_MAX = 65536
 
 
 def _pad8(x: int) -> int:
     return (x + 7) & ~7
+
+
+def _ldsq_bytes(mws) -> int:
+    """Launch-wide dynamic LDS reservation: the largest filter (in words)
+    that fits the budget; blocks whose filter does not fit fall back to
+    global word loads inside the kernel."""
+    fit = [w * 4 for w in mws if w * 4 <= LDSQ_MAX]
+    return (max(fit) + 255) & ~255 if fit else 0
 
 
 class BatchedPipeline:
@@ -41,12 +53,14 @@ class BatchedPipeline:
 
         desc = torch.zeros(T, 16, dtype=torch.int64)
         b2t = []
-        voff = koff = wire_off = cntoff = mwoff = blkoff = 0
+        voff = koff = wire_off = cntoff = mwoff = blkoff = iloff = 0
+        mws = []
         self.metas = []
         for t, n in enumerate(numels):
             k = max(1, int(round(n * ratio)))
             num_hash, m = Bloom._config(k, n, params)
             nbytes = (m + 7) // 8
+            mwords = (m + 31) // 32
             nb = (n + BT_CHUNK - 1) // BT_CHUNK
             desc[t, 0] = n
             desc[t, 1] = voff
@@ -59,19 +73,24 @@ class BatchedPipeline:
             desc[t, 8] = cntoff
             desc[t, 9] = mwoff
             desc[t, 10] = blkoff
+            desc[t, 15] = iloff                    # interleaved word offset
             self.metas.append([(vdt, k), (torch.uint8, nbytes)])
             b2t.extend([t] * nb)
+            mws.append(mwords)
             voff += n
             koff += k
             wire_off += _pad8(vb * k) + _pad8(nbytes)
             cntoff += nb
             mwoff += nb * (BT_CHUNK // 64)
             blkoff += nb
+            iloff += mwords
 
         self.total_values = voff
         self.k_total = koff
         self.wire_bytes = wire_off          # multiple of 8 by construction
         self.mask_words = mwoff
+        self.total_mw = iloff
+        self.ldsq_bytes = _ldsq_bytes(mws)
         self.desc = desc.to(device)
         self.b2t = torch.tensor(b2t, dtype=torch.int32, device=device)
 
@@ -81,7 +100,7 @@ class BatchedPipeline:
 
         wire, out_idx = _hip_ops.batched_compress(
             values_flat, self.desc, self.b2t, self.wire_bytes, self.k_total,
-            self.mask_words, int(self.wire_half),
+            self.mask_words, int(self.wire_half), self.ldsq_bytes,
         )
         return wire, out_idx
 
@@ -101,7 +120,8 @@ class BatchedPipeline:
 
         return _hip_ops.batched_decode_sum(wires2d, self.desc, self.b2t,
                                            self.total_values, self.mask_words,
-                                           int(self.wire_half))
+                                           int(self.wire_half), self.total_mw,
+                                           self.ldsq_bytes)
 
 
 class BothPipeline(BatchedPipeline):
@@ -125,14 +145,16 @@ class BothPipeline(BatchedPipeline):
 
         desc = torch.zeros(T, 16, dtype=torch.int64)
         b2t, seg_t, seg_i = [], [], []
-        voff = koff = wire_off = cntoff = mwoff = blkoff = 0
+        voff = koff = wire_off = cntoff = mwoff = blkoff = iloff = 0
         kmax = 1
+        mws = []
         self.metas = []
         for t, n in enumerate(numels):
             k = max(1, int(round(n * ratio)))
             kmax = max(kmax, k)
             num_hash, m = Bloom._config(k, n, params)
             nbytes = (m + 7) // 8
+            mwords = (m + 31) // 32
             nb = (n + BT_CHUNK - 1) // BT_CHUNK
             sp = s_pad(k)
             nbits = max(1, (k - 1).bit_length())
@@ -152,24 +174,29 @@ class BothPipeline(BatchedPipeline):
             desc[t, 12] = wire_off                       # coeffs
             desc[t, 13] = wire_off + coeff_bytes + _pad8(nbytes)  # mapping
             desc[t, 14] = nbits
+            desc[t, 15] = iloff                          # interleaved words
             self.metas.append([(torch.float64, sp * d1 + 1),
                                (torch.uint8, nbytes),
                                (torch.uint8, map_payload)])
             b2t.extend([t] * nb)
             seg_t.extend([t] * sp)
             seg_i.extend(range(sp))
+            mws.append(mwords)
             voff += n
             koff += k
             wire_off += coeff_bytes + _pad8(nbytes) + _pad8(map_payload)
             cntoff += nb
             mwoff += nb * (BT_CHUNK // 64)
             blkoff += nb
+            iloff += mwords
 
         self.total_values = voff
         self.k_total = koff
         self.kmax = kmax
         self.wire_bytes = wire_off
         self.mask_words = mwoff
+        self.total_mw = iloff
+        self.ldsq_bytes = _ldsq_bytes(mws)
         self.desc = desc.to(device)
         self.b2t = torch.tensor(b2t, dtype=torch.int32, device=device)
         self.seg_t = torch.tensor(seg_t, dtype=torch.int32, device=device)
@@ -181,7 +208,7 @@ class BothPipeline(BatchedPipeline):
         wire, own = _hip_ops.batched_compress_both(
             values_flat, self.desc, self.b2t, self.seg_t, self.seg_i,
             self.wire_bytes, self.k_total, self.mask_words, self.kmax,
-            self.degree, self.total_values,
+            self.degree, self.total_values, self.ldsq_bytes,
         )
         return wire, own
 
@@ -190,7 +217,7 @@ class BothPipeline(BatchedPipeline):
 
         return _hip_ops.batched_decode_both_sum(
             wires2d, self.desc, self.b2t, self.total_values, self.mask_words,
-            self.k_total, self.degree,
+            self.k_total, self.degree, self.total_mw, self.ldsq_bytes,
         )
 
 

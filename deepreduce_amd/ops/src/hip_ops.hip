@@ -1152,13 +1152,23 @@ __global__ void bt_insert_kernel(const int64_t* __restrict__ out_i,
 }
 
 // universe query over R wire buffers (filters share probe positions):
-// counts plane per rank + ballot bit-plane per rank
+// counts plane per rank + ballot bit-plane per rank.
+//
+// Memory path: probes are 32-bit WORD loads (one line
+// touch per probe, not a byte gather), and for R == 1 (the compress-side
+// query — 43% of compression cycles in profiles/r08) the whole filter is
+// staged into LDS once per block when it fits the launch's dynamic share
+// (ldsq_words; flagship filters are 10-45 KB): the k probes then hit LDS
+// at ~50 cyc instead of L1/L2 at 180+.  Multi-rank decode uses the
+// interleaved variant below instead.
 __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
                                  int64_t wstride, int R,
                                  const int64_t* __restrict__ desc,
                                  const int* __restrict__ b2t, int64_t BV,
-                                 int64_t MW, int* __restrict__ qcounts,
+                                 int64_t MW, int ldsq_words,
+                                 int* __restrict__ qcounts,
                                  uint64_t* __restrict__ mask) {
+    extern __shared__ uint32_t ldsq[];
     const int t = b2t[blockIdx.x];
     const int64_t* D = bt_row(desc, t);
     const int64_t lb = blockIdx.x - D[10];
@@ -1168,6 +1178,14 @@ __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
     const int nh = (int)D[5];
     const int64_t bitoff = D[6];
     const int64_t mwoff = D[9];
+    const int64_t mw = (m + 31) >> 5;
+    const uint32_t* __restrict__ w32 = (const uint32_t*)(wires + bitoff);
+    const bool use_lds = (R == 1) && (mw <= (int64_t)ldsq_words);
+    if (use_lds) {
+        for (int64_t w = threadIdx.x; w < mw; w += blockDim.x)
+            ldsq[w] = w32[w];
+        __syncthreads();
+    }
     int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
     int cnt[MAXR];
     for (int r = 0; r < R; ++r) cnt[r] = 0;
@@ -1180,12 +1198,100 @@ __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
             uint64_t pos = (uint64_t)h1 % (uint64_t)m;
             uint64_t step = (uint64_t)h2 % (uint64_t)m;
             for (int h = 0; h < nh && alive; ++h) {
-                int64_t byte = pos >> 3;
-                uint8_t bit = pos & 7;
+                const int64_t word = pos >> 5;
+                const uint32_t bit = 1u << (pos & 31);
+                if (use_lds) {
+                    if (!(ldsq[word] & bit)) alive = 0u;
+                } else {
+                    for (int r = 0; r < R; ++r)
+                        if (alive & (1u << r))
+                            if (!(((const uint32_t*)(wires + (int64_t)r * wstride
+                                                     + bitoff))[word] & bit))
+                                alive &= ~(1u << r);
+                }
+                pos += step;
+                if (pos >= (uint64_t)m) pos -= (uint64_t)m;
+            }
+        }
+        for (int r = 0; r < R; ++r) {
+            uint64_t ball = __ballot(alive & (1u << r));
+            cnt[r] += __popcll(ball);
+            if (lane == 0)
+                mask[r * MW + mwoff + ((i0 + (int64_t)wid * WAVE) >> 6)] = ball;
+        }
+    }
+    __shared__ int wsum[MAXR][QBLOCK / WAVE];
+    if (lane == 0)
+        for (int r = 0; r < R; ++r) wsum[r][wid] = cnt[r];
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int r = 0; r < R; ++r) {
+            int total = 0;
+            for (int w = 0; w < QBLOCK / WAVE; ++w) total += wsum[r][w];
+            qcounts[r * BV + D[8] + lb] = total;
+        }
+    }
+}
+
+// Interleave the R ranks' bloom bit arrays word-by-word:
+//   il[(iloff_t + w) * R + r] = word w of tensor t's filter in rank r's wire
+// so the decode-side probe of position pos touches R CONSECUTIVE words
+// (one or two cache lines) instead of R lines wstride apart.  One cheap
+// streaming pass (R * sum(mw) words) before the universe query.
+__global__ void bt_interleave_kernel(const uint8_t* __restrict__ wires,
+                                     int64_t wstride, int R,
+                                     const int64_t* __restrict__ desc, int nT,
+                                     int64_t total_mw,
+                                     uint32_t* __restrict__ il) {
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t total = total_mw * R;
+    for (; g < total; g += stride) {
+        const int64_t widx = g / R;
+        const int r = (int)(g - widx * R);
+        int lo = 0, hi = nT - 1;   // largest t with iloff[t] <= widx
+        while (lo < hi) {
+            int mid = (lo + hi + 1) >> 1;
+            if (bt_row(desc, mid)[15] <= widx) lo = mid; else hi = mid - 1;
+        }
+        const int64_t* D = bt_row(desc, lo);
+        const uint32_t* w32 = (const uint32_t*)(wires + (int64_t)r * wstride + D[6]);
+        il[g] = w32[widx - D[15]];
+    }
+}
+
+// Multi-rank universe query against the interleaved filter block.
+__global__ void bt_qcount_inter_kernel(const uint32_t* __restrict__ il, int R,
+                                       const int64_t* __restrict__ desc,
+                                       const int* __restrict__ b2t, int64_t BV,
+                                       int64_t MW, int* __restrict__ qcounts,
+                                       uint64_t* __restrict__ mask) {
+    const int t = b2t[blockIdx.x];
+    const int64_t* D = bt_row(desc, t);
+    const int64_t lb = blockIdx.x - D[10];
+    const int64_t start = lb * BT_CHUNK;
+    const int64_t end = min(start + BT_CHUNK, D[0]);
+    const int64_t m = D[4];
+    const int nh = (int)D[5];
+    const int64_t mwoff = D[9];
+    const uint32_t* __restrict__ ilt = il + D[15] * R;
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    int cnt[MAXR];
+    for (int r = 0; r < R; ++r) cnt[r] = 0;
+    for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
+        int64_t i = i0 + threadIdx.x;
+        unsigned alive = (i < end) ? ((1u << R) - 1) : 0u;
+        if (alive) {
+            uint32_t h1, h2;
+            hash_bases(i, &h1, &h2);
+            uint64_t pos = (uint64_t)h1 % (uint64_t)m;
+            uint64_t step = (uint64_t)h2 % (uint64_t)m;
+            for (int h = 0; h < nh && alive; ++h) {
+                const uint32_t bit = 1u << (pos & 31);
+                const uint32_t* __restrict__ row = ilt + (pos >> 5) * R;
                 for (int r = 0; r < R; ++r)
                     if (alive & (1u << r))
-                        if (!((wires[r * wstride + bitoff + byte] >> bit) & 1))
-                            alive &= ~(1u << r);
+                        if (!(row[r] & bit)) alive &= ~(1u << r);
                 pos += step;
                 if (pos >= (uint64_t)m) pos -= (uint64_t)m;
             }
@@ -1363,7 +1469,8 @@ std::vector<torch::Tensor> batched_compress(torch::Tensor values_flat,
                                             torch::Tensor b2t,
                                             int64_t wire_bytes, int64_t k_total,
                                             int64_t mask_words,
-                                            int64_t wire_half) {
+                                            int64_t wire_half,
+                                            int64_t ldsq_bytes) {
     CHECK_CUDA(values_flat);
     auto v = values_flat.contiguous();
     auto d = desc.contiguous();
@@ -1409,9 +1516,11 @@ std::vector<torch::Tensor> batched_compress(torch::Tensor values_flat,
     hipLaunchKernelGGL(bt_insert_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
                        out_idx.data_ptr<int64_t>(), dp, T, k_total,
                        wire.data_ptr<uint8_t>());
-    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK),
+                       (size_t)ldsq_bytes, stream,
                        wire.data_ptr<uint8_t>(), wire_bytes, 1, dp, mp, BV,
-                       mask_words, qcounts, (uint64_t*)mask.data_ptr<int64_t>());
+                       mask_words, (int)(ldsq_bytes / 4), qcounts,
+                       (uint64_t*)mask.data_ptr<int64_t>());
     hipLaunchKernelGGL(bt_scan_kernel, dim3(T), dim3(QBLOCK), 0, stream,
                        qcounts, dp, T, BV, qoffs);
     hipLaunchKernelGGL(bt_qscatter_own_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
@@ -1446,7 +1555,8 @@ torch::Tensor batched_scatter_dense(torch::Tensor wire, torch::Tensor out_idx,
 // sequentially so the accumulation order is deterministic on every rank.
 torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
                                  torch::Tensor b2t, int64_t total_values,
-                                 int64_t mask_words, int64_t wire_half) {
+                                 int64_t mask_words, int64_t wire_half,
+                                 int64_t total_mw, int64_t ldsq_bytes) {
     CHECK_CUDA(wires2d);
     TORCH_CHECK(wires2d.dim() == 2, "expected [R, W]");
     auto w = wires2d.contiguous();
@@ -1468,10 +1578,26 @@ torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
     auto dense = torch::empty({total_values}, torch::dtype(torch::kFloat32).device(dev));
     hipLaunchKernelGGL(bt_fill_zero_f, dim3(bt_grid(total_values)), dim3(256), 0, stream,
                        dense.data_ptr<float>(), total_values);
-    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
-                       w.data_ptr<uint8_t>(), W, R, d.data_ptr<int64_t>(),
-                       map.data_ptr<int>(), BV, mask_words, qcounts,
-                       (uint64_t*)mask.data_ptr<int64_t>());
+    if (R > 1 && total_mw > 0) {
+        auto il = torch::empty({(int64_t)R * total_mw},
+                               torch::dtype(torch::kInt32).device(dev));
+        hipLaunchKernelGGL(bt_interleave_kernel, dim3(bt_grid(total_mw * R)),
+                           dim3(256), 0, stream, w.data_ptr<uint8_t>(), W, R,
+                           d.data_ptr<int64_t>(), T, total_mw,
+                           (uint32_t*)il.data_ptr<int>());
+        hipLaunchKernelGGL(bt_qcount_inter_kernel, dim3((int)BV), dim3(QBLOCK), 0,
+                           stream, (const uint32_t*)il.data_ptr<int>(), R,
+                           d.data_ptr<int64_t>(), map.data_ptr<int>(), BV,
+                           mask_words, qcounts,
+                           (uint64_t*)mask.data_ptr<int64_t>());
+    } else {
+        hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK),
+                           (size_t)ldsq_bytes, stream,
+                           w.data_ptr<uint8_t>(), W, R, d.data_ptr<int64_t>(),
+                           map.data_ptr<int>(), BV, mask_words,
+                           (int)(ldsq_bytes / 4), qcounts,
+                           (uint64_t*)mask.data_ptr<int64_t>());
+    }
     hipLaunchKernelGGL(bt_scan_kernel, dim3(R * T), dim3(QBLOCK), 0, stream,
                        qcounts, d.data_ptr<int64_t>(), T, BV, qoffs);
     for (int r = 0; r < R; ++r) {
@@ -2029,7 +2155,7 @@ std::vector<torch::Tensor> batched_compress_both(
         torch::Tensor values_flat, torch::Tensor desc, torch::Tensor b2t,
         torch::Tensor seg_t, torch::Tensor seg_i, int64_t wire_bytes,
         int64_t k_total, int64_t mask_words, int64_t kmax, int64_t degree,
-        int64_t total_values) {
+        int64_t total_values, int64_t ldsq_bytes) {
     CHECK_CUDA(values_flat);
     auto v = values_flat.contiguous();
     auto d = desc.contiguous();
@@ -2083,9 +2209,11 @@ std::vector<torch::Tensor> batched_compress_both(
     hipLaunchKernelGGL(bt_insert_kernel, dim3(bt_grid(k_total)), dim3(256), 0, stream,
                        out_idx.data_ptr<int64_t>(), dp, T, k_total,
                        wire.data_ptr<uint8_t>());
-    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
+    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK),
+                       (size_t)ldsq_bytes, stream,
                        wire.data_ptr<uint8_t>(), wire_bytes, 1, dp, mp, BV,
-                       mask_words, qcounts, (uint64_t*)mask.data_ptr<int64_t>());
+                       mask_words, (int)(ldsq_bytes / 4), qcounts,
+                       (uint64_t*)mask.data_ptr<int64_t>());
     hipLaunchKernelGGL(bt_scan_kernel, dim3(T), dim3(QBLOCK), 0, stream,
                        qcounts, dp, T, BV, qoffs);
     // FP-aware gather to the temp buffer; positives into out_idx
@@ -2132,7 +2260,8 @@ std::vector<torch::Tensor> batched_compress_both(
 torch::Tensor batched_decode_both_sum(torch::Tensor wires2d, torch::Tensor desc,
                                       torch::Tensor b2t, int64_t total_values,
                                       int64_t mask_words, int64_t k_total,
-                                      int64_t degree) {
+                                      int64_t degree, int64_t total_mw,
+                                      int64_t ldsq_bytes) {
     CHECK_CUDA(wires2d);
     TORCH_CHECK(wires2d.dim() == 2, "expected [R, W]");
     auto w = wires2d.contiguous();
@@ -2159,10 +2288,26 @@ torch::Tensor batched_decode_both_sum(torch::Tensor wires2d, torch::Tensor desc,
                                torch::dtype(torch::kInt64).device(dev));
     hipLaunchKernelGGL(bt_fill_zero_f, dim3(bt_grid(total_values)), dim3(256), 0, stream,
                        dense.data_ptr<float>(), total_values);
-    hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
-                       w.data_ptr<uint8_t>(), W, R, d.data_ptr<int64_t>(),
-                       map.data_ptr<int>(), BV, mask_words, qcounts,
-                       (uint64_t*)mask.data_ptr<int64_t>());
+    if (R > 1 && total_mw > 0) {
+        auto il = torch::empty({(int64_t)R * total_mw},
+                               torch::dtype(torch::kInt32).device(dev));
+        hipLaunchKernelGGL(bt_interleave_kernel, dim3(bt_grid(total_mw * R)),
+                           dim3(256), 0, stream, w.data_ptr<uint8_t>(), W, R,
+                           d.data_ptr<int64_t>(), T, total_mw,
+                           (uint32_t*)il.data_ptr<int>());
+        hipLaunchKernelGGL(bt_qcount_inter_kernel, dim3((int)BV), dim3(QBLOCK), 0,
+                           stream, (const uint32_t*)il.data_ptr<int>(), R,
+                           d.data_ptr<int64_t>(), map.data_ptr<int>(), BV,
+                           mask_words, qcounts,
+                           (uint64_t*)mask.data_ptr<int64_t>());
+    } else {
+        hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK),
+                           (size_t)ldsq_bytes, stream,
+                           w.data_ptr<uint8_t>(), W, R, d.data_ptr<int64_t>(),
+                           map.data_ptr<int>(), BV, mask_words,
+                           (int)(ldsq_bytes / 4), qcounts,
+                           (uint64_t*)mask.data_ptr<int64_t>());
+    }
     hipLaunchKernelGGL(bt_scan_kernel, dim3(R * T), dim3(QBLOCK), 0, stream,
                        qcounts, d.data_ptr<int64_t>(), T, BV, qoffs);
     for (int r = 0; r < R; ++r) {
@@ -2364,6 +2509,209 @@ torch::Tensor batched_decode_value_sum(torch::Tensor wires2d, torch::Tensor desc
 
 #include <ATen/Parallel.h>
 
+// ---------------------------------------------------------------------------
+// DoubleExp (Fit-DExp) fused fit kernel — VERDICT r1 item 6.
+//
+// Reference behavior: tensorflow/deepreduce.py:67-144 (Jacquelin cumulative-
+// integral linearization of y = a*e^{bx} + c*e^{dx}): regress y on
+// [SS, S, x, 1] where S = cumint(y), SS = cumint(S); b,d are the roots of
+// t^2 - B t - A; then a,c from the 2x2 LS of y on (e^{bx}, e^{dx}).
+// The torch path chained ~20 fp64 ops (two cumsums, stacks, matmuls, two
+// linalg.solve -> 2 hipSolver round trips); here ONE block per tensor does
+// both passes: chunked block-scan for S/SS fused with the Gram/moment
+// accumulation, in-register Cholesky for the 4x4 and 2x2 solves.
+// y must be sorted ascending (the codec sorts by |value|), x = i+1.
+// ---------------------------------------------------------------------------
+
+#define DEXP_BLOCK 256
+
+__device__ __forceinline__ double dexp_block_scan(double v, int wid, int lane,
+                                                  double* wtot, double* carry) {
+    // inclusive block scan of v (one value per thread); carry holds the
+    // running prefix from previous chunks and is updated by thread 0
+    double incl = v;
+    for (int off = 1; off < WAVE; off <<= 1) {
+        double up = __shfl_up(incl, off, WAVE);
+        if (lane >= off) incl += up;
+    }
+    if (lane == WAVE - 1) wtot[wid] = incl;
+    __syncthreads();
+    double wbase = *carry;
+    for (int w = 0; w < wid; ++w) wbase += wtot[w];
+    double out = wbase + incl;
+    __syncthreads();
+    if (threadIdx.x == blockDim.x - 1) *carry = out;
+    __syncthreads();
+    return out;
+}
+
+__device__ __forceinline__ void cholesky_solve_inreg(double* G, double* b, int n) {
+    // G [n,n] row-major SPD (ridged), solve G x = b in place -> b
+    for (int i = 0; i < n; ++i) {
+        for (int j_ = 0; j_ <= i; ++j_) {
+            double s = G[i * n + j_];
+            for (int p = 0; p < j_; ++p) s -= G[i * n + p] * G[j_ * n + p];
+            if (i == j_) G[i * n + i] = sqrt(fmax(s, 1e-300));
+            else G[i * n + j_] = s / G[j_ * n + j_];
+        }
+    }
+    for (int i = 0; i < n; ++i) {          // forward
+        double s = b[i];
+        for (int p = 0; p < i; ++p) s -= G[i * n + p] * b[p];
+        b[i] = s / G[i * n + i];
+    }
+    for (int i = n - 1; i >= 0; --i) {     // backward
+        double s = b[i];
+        for (int p = i + 1; p < n; ++p) s -= G[p * n + i] * b[p];
+        b[i] = s / G[i * n + i];
+    }
+}
+
+__global__ void dexp_fit_kernel(const float* __restrict__ y_flat,
+                                const int64_t* __restrict__ offs,
+                                const int64_t* __restrict__ lens,
+                                double* __restrict__ coeffs /*[B,4]*/) {
+    const int b_idx = blockIdx.x;
+    const float* __restrict__ y = y_flat + offs[b_idx];
+    const int64_t N = lens[b_idx];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+
+    __shared__ double wtot[DEXP_BLOCK / WAVE];
+    __shared__ double carrS, carrSS, prevy, prevS;
+    __shared__ double red[DEXP_BLOCK / WAVE][14];
+    __shared__ double theta_s[4];
+
+    if (threadIdx.x == 0) { carrS = 0.0; carrSS = 0.0; prevy = 0.0; prevS = 0.0; }
+    __syncthreads();
+
+    double acc[14];
+    for (int q = 0; q < 14; ++q) acc[q] = 0.0;
+
+    __shared__ double ylast[DEXP_BLOCK / WAVE];
+    __shared__ double Slast[DEXP_BLOCK / WAVE];
+    for (int64_t i0 = 0; i0 < N; i0 += blockDim.x) {
+        const int64_t i = i0 + threadIdx.x;
+        const bool ok = i < N;
+        const double yi = ok ? (double)y[i] : 0.0;
+        // neighbor y[i-1]: shfl within the wave, LDS across wave/chunk edges
+        if (lane == WAVE - 1) ylast[wid] = yi;
+        __syncthreads();
+        const double upy = __shfl_up(yi, 1, WAVE);
+        const double ym1 = (lane > 0) ? upy
+                           : ((wid == 0) ? prevy : ylast[wid - 1]);
+        const double ds = (i == 0 || !ok) ? 0.0 : 0.5 * (yi + ym1);
+        __syncthreads();  // prevy read by all before the update below
+        if (ok && threadIdx.x == blockDim.x - 1) prevy = yi;
+        const double Si = dexp_block_scan(ds, wid, lane, wtot, &carrS);
+
+        // second trapezoid: (S[i]+S[i-1])/2
+        if (lane == WAVE - 1) Slast[wid] = Si;
+        __syncthreads();
+        const double upS = __shfl_up(Si, 1, WAVE);
+        const double Sm1 = (lane > 0) ? upS
+                           : ((wid == 0) ? prevS : Slast[wid - 1]);
+        const double dss = (i == 0 || !ok) ? 0.0 : 0.5 * (Si + Sm1);
+        __syncthreads();
+        if (ok && threadIdx.x == blockDim.x - 1) prevS = Si;
+        const double SSi = dexp_block_scan(dss, wid, lane, wtot, &carrSS);
+
+        if (ok) {
+            const double xi = (double)(i + 1);
+            // Gram of [SS, S, x, 1] (10 upper entries) + rhs (4)
+            acc[0] += SSi * SSi; acc[1] += SSi * Si; acc[2] += SSi * xi;
+            acc[3] += SSi;       acc[4] += Si * Si;  acc[5] += Si * xi;
+            acc[6] += Si;        acc[7] += xi * xi;  acc[8] += xi;
+            acc[9] += 1.0;
+            acc[10] += SSi * yi; acc[11] += Si * yi; acc[12] += xi * yi;
+            acc[13] += yi;
+        }
+        __syncthreads();
+    }
+
+    // block-reduce the 14 accumulators
+    for (int q = 0; q < 14; ++q) {
+        double v = acc[q];
+        for (int off = WAVE / 2; off > 0; off >>= 1)
+            v += __shfl_down(v, off, WAVE);
+        if (lane == 0) red[wid][q] = v;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        double G[16], rhs[4];
+        double s[14];
+        for (int q = 0; q < 14; ++q) {
+            double v = 0.0;
+            for (int w = 0; w < DEXP_BLOCK / WAVE; ++w) v += red[w][q];
+            s[q] = v;
+        }
+        G[0] = s[0];  G[1] = s[1];  G[2] = s[2];  G[3] = s[3];
+        G[4] = s[1];  G[5] = s[4];  G[6] = s[5];  G[7] = s[6];
+        G[8] = s[2];  G[9] = s[5];  G[10] = s[7]; G[11] = s[8];
+        G[12] = s[3]; G[13] = s[6]; G[14] = s[8]; G[15] = s[9];
+        rhs[0] = s[10]; rhs[1] = s[11]; rhs[2] = s[12]; rhs[3] = s[13];
+        double mx = 0.0;
+        for (int q = 0; q < 4; ++q) mx = fmax(mx, fabs(G[q * 4 + q]));
+        const double ridge = mx * 1e-12 + 1e-30;
+        for (int q = 0; q < 4; ++q) G[q * 4 + q] += ridge;
+        cholesky_solve_inreg(G, rhs, 4);
+        const double A = rhs[0], B = rhs[1];
+        const double disc = fmax(B * B + 4.0 * A, 0.0);
+        const double r = sqrt(disc);
+        const double cap = 650.0 / fmax((double)N, 1.0);
+        theta_s[0] = fmin(fmax(0.5 * (B + r), -cap), cap);   // b
+        theta_s[1] = fmin(fmax(0.5 * (B - r), -cap), cap);   // d
+    }
+    __syncthreads();
+    const double bb = theta_s[0], dd = theta_s[1];
+
+    // pass 2: 2x2 LS of y on (e^{b x}, e^{d x})
+    double g2[5] = {0, 0, 0, 0, 0};  // ebeb, ebed, eded, eby, edy
+    for (int64_t i = threadIdx.x; i < N; i += blockDim.x) {
+        const double xi = (double)(i + 1);
+        const double yi = (double)y[i];
+        const double eb = exp(bb * xi), ed = exp(dd * xi);
+        g2[0] += eb * eb; g2[1] += eb * ed; g2[2] += ed * ed;
+        g2[3] += eb * yi; g2[4] += ed * yi;
+    }
+    for (int q = 0; q < 5; ++q) {
+        double v = g2[q];
+        for (int off = WAVE / 2; off > 0; off >>= 1)
+            v += __shfl_down(v, off, WAVE);
+        if (lane == 0) red[wid][q] = v;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        double s2[5];
+        for (int q = 0; q < 5; ++q) {
+            double v = 0.0;
+            for (int w = 0; w < DEXP_BLOCK / WAVE; ++w) v += red[w][q];
+            s2[q] = v;
+        }
+        double G2[4] = {s2[0] + 1e-12, s2[1], s2[1], s2[2] + 1e-12};
+        double ac[2] = {s2[3], s2[4]};
+        cholesky_solve_inreg(G2, ac, 2);
+        coeffs[(int64_t)b_idx * 4 + 0] = ac[0];
+        coeffs[(int64_t)b_idx * 4 + 1] = bb;
+        coeffs[(int64_t)b_idx * 4 + 2] = ac[1];
+        coeffs[(int64_t)b_idx * 4 + 3] = dd;
+    }
+}
+
+torch::Tensor dexp_fit(torch::Tensor y, torch::Tensor offs, torch::Tensor lens) {
+    CHECK_CUDA(y);
+    auto yy = y.contiguous();
+    auto o = offs.contiguous();
+    auto l = lens.contiguous();
+    TORCH_CHECK(yy.dtype() == torch::kFloat32, "y must be float32");
+    const int B = (int)o.numel();
+    auto out = torch::empty({B, 4}, torch::dtype(torch::kFloat64).device(y.device()));
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(dexp_fit_kernel, dim3(B), dim3(DEXP_BLOCK), 0, stream,
+                       yy.data_ptr<float>(), o.data_ptr<int64_t>(),
+                       l.data_ptr<int64_t>(), out.data_ptr<double>());
+    return out;
+}
+
 torch::Tensor bloom_insert_cpu(torch::Tensor idxs, int64_t m, int64_t num_hash) {
     auto items = idxs.to(torch::kInt64).contiguous();
     int64_t nbytes = ceil_div(m, 8);
@@ -2546,6 +2894,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "whole-model value-mode compress: polyfit coeffs + int32 idxs");
     m.def("batched_decode_value_sum", &batched_decode_value_sum,
           "multi-rank value-mode decode");
+    m.def("dexp_fit", &dexp_fit,
+          "fused DoubleExp cumulative-integral fit (one block per tensor)");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");
     m.def("bloom_query_positives_cpu", &bloom_query_positives_cpu, "Bloom query (C++ CPU)");
     m.def("bloom_query_members_cpu", &bloom_query_members_cpu, "Bloom members (C++ CPU)");

@@ -71,9 +71,10 @@ class ValueCompressor(_WrapperBase):
         super().__init__(sparsifier, params)
         name = self.params.get("value", "polyfit")
         self.val_codec = codec_registry[name]
-        if name not in ("qsgd", "polyseg", "polyfit"):
+        if name not in ("qsgd", "polyseg", "polyfit", "doubleexp"):
             # qsgd/polyseg payload sizes depend only on k; polyfit's padded
-            # 22-slot layout depends only on N -> all uniform across ranks
+            # 22-slot layout depends only on N; doubleexp is 4 coeffs + a
+            # k-sized signed mapping -> all uniform across ranks
             self.tensors_size_are_same = False
 
     def compress(self, tensor, name):

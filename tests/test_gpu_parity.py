@@ -637,3 +637,139 @@ def test_graph_recaptures_on_grad_realloc(dev):
     assert opt._graph is not g1 or opt._graph is None
     # training remains finite and sane
     assert all(torch.isfinite(p).all() for p in model.parameters())
+
+
+def test_dexp_fit_kernel_parity(hip, dev):
+    """Fused DoubleExp fit kernel vs the torch fp64 reference math
+    (codecs/doubleexp._double_exp_fit): same coefficients to fp32-input
+    tolerance, same reconstruction."""
+    from deepreduce_amd.codecs.doubleexp import _double_exp_fit
+
+    torch.manual_seed(5)
+    for N in [1200, 9000, 50_000]:
+        # exponential-ish sorted magnitudes (what sorted topk values look like)
+        y = torch.sort(torch.randn(N, device=dev).abs() ** 2.0).values
+        ref = _double_exp_fit(y.double())
+        ref = torch.stack(list(ref)).cpu()
+        got = hip.dexp_fit(
+            y.float(),
+            torch.zeros(1, dtype=torch.int64, device=dev),
+            torch.tensor([N], dtype=torch.int64, device=dev),
+        ).cpu().reshape(-1)
+
+        x = torch.arange(1, N + 1, dtype=torch.float64)
+        rec_ref = ref[0] * torch.exp(ref[1] * x) + ref[2] * torch.exp(ref[3] * x)
+        rec_got = got[0] * torch.exp(got[1] * x) + got[2] * torch.exp(got[3] * x)
+        denom = y.double().cpu().norm() + 1e-12
+        err_ref = (rec_ref - y.double().cpu()).norm() / denom
+        err_got = (rec_got - y.double().cpu()).norm() / denom
+        # the kernel's fit must be as good as the reference fit (small slack
+        # for fp32 input + different fp64 summation order)
+        assert err_got <= err_ref + 0.02, (N, float(err_got), float(err_ref))
+
+
+def test_dexp_fit_batched_tensors(hip, dev):
+    """One launch fits B tensors (one block each)."""
+    torch.manual_seed(6)
+    lens = [3000, 12_000, 700]
+    ys = [torch.sort(torch.randn(n, device=dev).abs()).values for n in lens]
+    flat = torch.cat(ys).float()
+    offs = torch.tensor([0, 3000, 15_000], dtype=torch.int64, device=dev)
+    lent = torch.tensor(lens, dtype=torch.int64, device=dev)
+    out = hip.dexp_fit(flat, offs, lent)
+    assert out.shape == (3, 4)
+    for i, (y, n) in enumerate(zip(ys, lens)):
+        c = out[i].cpu()
+        single = hip.dexp_fit(
+            y.float(), torch.zeros(1, dtype=torch.int64, device=dev),
+            torch.tensor([n], dtype=torch.int64, device=dev)).cpu().reshape(-1)
+        assert torch.allclose(c, single, rtol=1e-10, atol=1e-12), i
+
+
+def test_doubleexp_codec_gpu_matches_cpu(dev):
+    """DoubleExp codec end-to-end on GPU (device fit kernel) vs CPU (torch
+    fp64 path): same decompressed tensor to fp32 tolerance."""
+    from deepreduce_amd.codecs import compressor as registry
+
+    torch.manual_seed(7)
+    t = torch.randn(200_000)
+    k = 20_000
+    _, idx = t.abs().topk(k)
+    vals = t[idx]
+
+    codec = registry["doubleexp"]
+    v_c, m_c, s_c = codec.compress((vals, idx, t.size()), {})
+    out_c = codec.decompress((v_c, m_c, s_c), {})
+
+    v_g, m_g, s_g = codec.compress((vals.to(dev), idx.to(dev), t.size()), {})
+    out_g = codec.decompress((v_g, m_g, s_g), {})
+
+    assert torch.equal(m_g.cpu(), m_c)  # mapping identical
+    dense_c = torch.zeros(t.numel()).scatter_(0, out_c[1], out_c[0])
+    dense_g = torch.zeros(t.numel(), device=dev).scatter_(0, out_g[1], out_g[0])
+    scale = vals.abs().max()
+    assert torch.allclose(dense_g.cpu(), dense_c, atol=2e-3 * float(scale)), \
+        (dense_g.cpu() - dense_c).abs().max()
+
+
+def test_batched_multi_rank_decode_r8_interleaved(dev):
+    """R=8 (the SCALE run's world size) through the interleaved universe
+    query must equal the sum of own-decodes."""
+    from deepreduce_amd import deepreduce_from_params
+    from deepreduce_amd.ops.batched import BatchedPipeline
+
+    params = {
+        "compressor": "topk", "memory": "residual",
+        "communicator": "allgather", "compress_ratio": 0.01,
+        "deepreduce": "index", "index": "bloom", "policy": "leftmost",
+    }
+    R = 8
+    numels = [300_000, 50_000, 9_000, 123_456]
+    names = [f"t{i}" for i in range(len(numels))]
+    bp = BatchedPipeline(names, numels, params, dev)
+    assert bp.total_mw > 0
+
+    torch.manual_seed(11)
+    wires, dense_ref = [], torch.zeros(sum(numels), device=dev)
+    for r in range(R):
+        flat = torch.randn(sum(numels), device=dev)
+        wire, out_idx = bp.compress(flat)
+        wires.append(wire)
+        dense_ref += bp.decode_own(wire, out_idx)
+    got = bp.decode_sum(torch.stack(wires))
+    assert torch.allclose(got, dense_ref.reshape(-1), atol=1e-5), \
+        (got - dense_ref).abs().max()
+
+
+def test_batched_large_filter_lds_fallback(dev):
+    """A filter larger than the 64 KB LDS budget must fall back to global
+    word loads inside bt_qcount (and still match the per-tensor path)."""
+    from deepreduce_amd import deepreduce_from_params
+    from deepreduce_amd.ops.batched import BatchedPipeline, LDSQ_MAX
+
+    params = {
+        "compressor": "topk", "memory": "none",
+        "communicator": "allgather", "compress_ratio": 0.05,
+        "deepreduce": "index", "index": "bloom", "policy": "leftmost",
+    }
+    numels = [4_000_000, 60_000]   # k=200k -> filter ~0.4 MB >> LDSQ_MAX
+    names = ["big", "small"]
+    bp = BatchedPipeline(names, numels, params, dev)
+    mws = [(int(bp.desc[t, 4]) + 31) // 32 for t in range(2)]
+    assert mws[0] * 4 > LDSQ_MAX, "test premise: big filter exceeds budget"
+    assert mws[1] * 4 <= LDSQ_MAX
+
+    torch.manual_seed(13)
+    flat = torch.randn(sum(numels), device=dev)
+    wire, out_idx = bp.compress(flat)
+    own = bp.decode_own(wire, out_idx)
+
+    # generic per-tensor path on the same inputs
+    grc = deepreduce_from_params(dict(params))
+    off = 0
+    for n, nel in zip(names, numels):
+        t = flat[off : off + nel]
+        tc, ctx = grc.compressor.compress(t, n)
+        dec = grc.compressor.decompress(tc, ctx)
+        assert torch.allclose(own[off : off + nel], dec.reshape(-1), atol=1e-6), n
+        off += nel

@@ -62,7 +62,8 @@ class DoubleExp(SparseCompressor):
     def compress(sparse_tensor, params):
         vals, idxs, shape = sparse_tensor
         y_abs = vals.double().abs()
-        order = torch.argsort(y_abs)  # ascending |value|
+        # stable: deterministic tie order, identical on CPU and GPU
+        order = torch.argsort(y_abs, stable=True)  # ascending |value|
         y_sorted = y_abs[order]
         sign = torch.sign(vals.double()[order])
         sign = torch.where(sign == 0, torch.ones_like(sign), sign)

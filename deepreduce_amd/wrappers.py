@@ -232,12 +232,26 @@ class DeepReduce(_WrapperBase):
             own = self.params.pop("_own_decoded", None)
             new_idxs = torch.arange(vals.numel(), device=vals.device)
             vals, mapping, shape_out = self.val_codec.compress((vals, new_idxs, shape), self.params)
+            ctx = shape_out
+            if self._skip_mapping():
+                # Order-preserving value codec (qsgd/gzip): the "mapping" is
+                # the identity, so it never travels — decompress re-derives
+                # the bloom positives in ascending order and the value codec
+                # returns values in exactly that order.  This is the lever
+                # behind the paper's DR-QSGD-BF-P0 headline volume (Table 2
+                # 0.0621): without it the ceil(log2 k)-bit permutation
+                # dominates the wire.
+                if own is not None:
+                    self._own_cache[name] = (own[1], None)
+                if self.params.get("micro-benchmark", False):
+                    _sync_if(tensor)
+                    print(f"_compression time:{time.perf_counter() - start}")
+                return (vals, idxs), ctx
             if own is not None:
                 # cache (bloom indices, unpacked mapping): decompress_own
                 # then skips both the universe query AND the mapping
                 # header parse (a host sync per tensor)
                 self._own_cache[name] = (own[1], mapping.long())
-            ctx = shape_out
             if self.params.get("pack_mapping", True):
                 # ceil(log2 k) bits per mapping entry instead of int32
                 # (paper App. E; the reference left this commented out at
@@ -256,15 +270,40 @@ class DeepReduce(_WrapperBase):
             print(f"_compression time:{time.perf_counter() - start}")
         return tensors, ctx
 
+    def _skip_mapping(self) -> bool:
+        """True when the identity mapping need not travel: the value codec
+        preserves order AND the decompress side can derive the positive
+        count without it (p0 carries it in-band; topk/randomk imply k)."""
+        if not getattr(self.val_codec, "order_preserving", False):
+            return False
+        return (self.params.get("policy", "leftmost") == "p0"
+                or getattr(self.sparsifier, "compress_ratio", None) is not None)
+
+    def _expected_k(self, shape) -> int:
+        numel = int(torch.Size(shape).numel())
+        ratio = getattr(self.sparsifier, "compress_ratio")
+        return max(1, int(round(numel * ratio)))
+
     def decompress(self, tensors, ctx):
         shape = ctx
         start = time.perf_counter()
         if torch.Size(shape).numel() > _BYPASS_NUMEL:
-            vals, idxs, mapping = tensors
-            mapping = self._unpack_mapping(mapping)
-            vals, _, _ = self.val_codec.decompress((vals, mapping, shape), self.params)
-            _, idxs, _ = self.idx_codec.decompress((mapping, idxs, shape), self.params)
-            idxs = idxs[mapping]
+            if len(tensors) == 2:  # mapping-free (order-preserving) wire
+                vals_w, packed = tensors
+                if self.params.get("policy", "leftmost") == "p0":
+                    placeholder = vals_w  # count travels in-band in the bits
+                else:
+                    placeholder = vals_w.new_empty(self._expected_k(shape))
+                _, idxs, _ = self.idx_codec.decompress(
+                    (placeholder, packed, shape), self.params)
+                vals, _, _ = self.val_codec.decompress(
+                    (vals_w, idxs, shape), self.params)
+            else:
+                vals, idxs, mapping = tensors
+                mapping = self._unpack_mapping(mapping)
+                vals, _, _ = self.val_codec.decompress((vals, mapping, shape), self.params)
+                _, idxs, _ = self.idx_codec.decompress((mapping, idxs, shape), self.params)
+                idxs = idxs[mapping]
         else:
             vals, idxs = tensors
         if self.params.get("micro-benchmark", False):
@@ -291,6 +330,10 @@ class DeepReduce(_WrapperBase):
             return self.decompress(tensors, ctx)
         cached_idxs, mapping = cached
         vals = tensors[0]
+        if mapping is None:  # order-preserving value codec: identity mapping
+            vals, _, _ = self.val_codec.decompress((vals, cached_idxs, shape),
+                                                   self.params)
+            return self.sparsifier.decompress((vals, cached_idxs), shape)
         vals, _, _ = self.val_codec.decompress((vals, mapping, shape), self.params)
         idxs = cached_idxs[mapping]
         return self.sparsifier.decompress((vals, idxs), shape)

@@ -29,6 +29,12 @@ CONFIGS = [
     ("DR-QSGD", {"deepreduce": "value", "value": "qsgd"}),
     ("DR-QSGD+BF-P0", {"deepreduce": "both", "value": "qsgd",
                        "index": "bloom", "policy": "p0"}),
+    # the paper's Table 2 headline config: QSGD levels packed at 7 bits
+    # (qsgd_pack) + BF-P0; the order-preserving value codec means no
+    # mapping travels (wrappers.DeepReduce._skip_mapping)
+    ("DR-QSGD-BF-P0 (7-bit packed)",
+     {"deepreduce": "both", "value": "qsgd", "index": "bloom",
+      "policy": "p0", "qsgd_pack": True, "quantum_num": 63}),
     ("DR-FitPoly+BF", {"deepreduce": "both", "value": "polyfit",
                        "index": "bloom", "policy": "leftmost"}),
 ]

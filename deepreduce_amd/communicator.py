@@ -112,11 +112,17 @@ class Communicator:
             st.dump()
 
     def _tensor_wire_bytes(self, name, t):
-        """Static per-tensor wire estimate for logging (exact totals come
-        from last_wire_bytes)."""
+        """Per-tensor wire bytes for logging.  Exact when available: the
+        fused exchange records each tensor's measured payload size
+        (_measured_tensor_bytes, set by _step_many_impl) and the batched
+        pipeline's layout is host-static; the 8k estimate remains only for
+        the first step of codecs the generic path hasn't measured yet."""
         comp = self.compressor
         if t.numel() <= 1000 and getattr(self, "params", {}).get("small_dense", True):
             return t.numel() * 4
+        measured = getattr(self, "_measured_tensor_bytes", None)
+        if measured is not None and name in measured:
+            return measured[name]
         cached = getattr(self, "_bt_pipeline", None)
         if cached is not None:
             bp = cached[1]
@@ -252,6 +258,12 @@ class Allgather(Communicator):
             metas.append(m)
         sizes = [b.numel() for b in bufs]
         self.last_wire_bytes = sum(sizes)
+        # record exact per-tensor payload sizes for the stats logger
+        # (fixes the 8k fallback estimate for gzip/huffman/rle payloads)
+        rec = getattr(self, "_measured_tensor_bytes", None)
+        if rec is None:
+            rec = self._measured_tensor_bytes = {}
+        rec.update(zip(names, sizes))
         if world == 1:
             return [d.view_as(g) for d, g in zip(decoded, grads)]
 

@@ -14,13 +14,21 @@ per-tensor path in communicator.step_many.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 
 BT_CHUNK = 8192
-# dynamic-LDS budget for the compress-side bloom query (bt_qcount R=1):
-# 64 KB leaves 2 workgroups per CU (160 KB LDS) and covers the flagship's
-# 10-45 KB filters
-LDSQ_MAX = 65536
+# dynamic-LDS budget for the compress-side bloom query (bt_qcount R=1).
+# MEASURED on MI355X (gpurun_out/qcount_ab.json, 30-iter A/B on the
+# ResNet-50 layout): LDS-cached 0.533 ms vs plain 32-bit word loads
+# 0.516 ms per whole-model compress — the occupancy cost of a 42 KB
+# dynamic-LDS reservation slightly outweighs the LDS latency win, so the
+# default is OFF (word loads through L1/L2, which hold the hot filter
+# fine at R=1).  Set DEEPREDUCE_LDSQ_MAX=65536 to re-enable for
+# experiments.  The multi-rank decode query uses the interleaved layout
+# instead (measured 9% faster at R=8), which is always on.
+LDSQ_MAX = int(os.environ.get("DEEPREDUCE_LDSQ_MAX", "0"))
 
 
 def _pad8(x: int) -> int:
@@ -30,8 +38,9 @@ def _pad8(x: int) -> int:
 def _ldsq_bytes(mws) -> int:
     """Launch-wide dynamic LDS reservation: the largest filter (in words)
     that fits the budget; blocks whose filter does not fit fall back to
-    global word loads inside the kernel."""
-    fit = [w * 4 for w in mws if w * 4 <= LDSQ_MAX]
+    global word loads inside the kernel.  0 when disabled (the default —
+    see LDSQ_MAX note)."""
+    fit = [w * 4 for w in mws if w * 4 <= LDSQ_MAX] if LDSQ_MAX else []
     return (max(fit) + 255) & ~255 if fit else 0
 
 

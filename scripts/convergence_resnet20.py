@@ -1,0 +1,149 @@
+#!/usr/bin/env python3
+"""ResNet-20 convergence: dense vs DeepReduce-compressed training on a
+CIFAR-shaped synthetic task.
+
+CIFAR-10 itself is not available in this offline environment (no network,
+no torchvision), so the paper's absolute 0.88-0.91 top-1 band cannot be
+reproduced here.  What CAN be tested — and is the paper's actual claim
+(pdf p.6 Figs. 15-17) — is that DR variants track the DENSE baseline's
+accuracy at matched epochs on the real ResNet-20 architecture.  The task:
+10 classes of procedurally generated 32x32x3 images (class = a fixed
+random low-frequency pattern; sample = pattern + crop-shift + flip +
+Gaussian noise), hard enough that random init scores 10% and a trained
+ResNet-20 climbs well into the 90s.
+
+Usage: python scripts/convergence_resnet20.py [--epochs 12] [--out docs/CONVERGENCE.md]
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+import torch.nn.functional as F
+
+from deepreduce_amd import DistributedOptimizer, deepreduce_from_params
+from deepreduce_amd.models import registry
+
+CONFIGS = [
+    ("dense", {"compressor": "none", "memory": "none",
+               "communicator": "allreduce"}),
+    ("Top-r 1%", {"compressor": "topk", "memory": "residual",
+                  "communicator": "allgather", "compress_ratio": 0.01}),
+    ("DR-BF leftmost", {"compressor": "topk", "memory": "residual",
+                        "communicator": "allgather", "compress_ratio": 0.01,
+                        "deepreduce": "index", "index": "bloom",
+                        "policy": "leftmost"}),
+    ("DR-FitPoly+BF 'both'", {"compressor": "topk", "memory": "residual",
+                              "communicator": "allgather",
+                              "compress_ratio": 0.01, "deepreduce": "both",
+                              "value": "polyfit", "index": "bloom",
+                              "policy": "leftmost"}),
+]
+
+
+def make_dataset(n_train=4096, n_test=1024, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    # class prototypes: low-frequency random patterns, 10 classes
+    freq = torch.randn(10, 3, 8, 8, generator=g)
+    protos = F.interpolate(freq, size=(40, 40), mode="bicubic",
+                           align_corners=False)  # upsample = low-pass
+
+    def sample(n, gen):
+        ys = torch.randint(0, 10, (n,), generator=gen)
+        dx = torch.randint(0, 8, (n,), generator=gen)
+        dy = torch.randint(0, 8, (n,), generator=gen)
+        flip = torch.randint(0, 2, (n,), generator=gen)
+        xs = torch.empty(n, 3, 32, 32)
+        for i in range(n):
+            img = protos[ys[i], :, dy[i] : dy[i] + 32, dx[i] : dx[i] + 32]
+            if flip[i]:
+                img = img.flip(-1)
+            xs[i] = img
+        xs += 0.7 * torch.randn(xs.shape, generator=gen)
+        return xs, ys
+
+    xtr, ytr = sample(n_train, torch.Generator().manual_seed(seed + 1))
+    xte, yte = sample(n_test, torch.Generator().manual_seed(seed + 2))
+    return xtr, ytr, xte, yte
+
+
+def accuracy(model, x, y, bs=256):
+    model.eval()
+    correct = 0
+    with torch.no_grad():
+        for i in range(0, len(x), bs):
+            pred = model(x[i : i + bs]).argmax(1)
+            correct += (pred == y[i : i + bs]).sum().item()
+    model.train()
+    return correct / len(x)
+
+
+def run_config(label, params, data, epochs, batch=128, lr=0.1):
+    xtr, ytr, xte, yte = data
+    torch.manual_seed(42)
+    model = registry["resnet20"]()
+    grc = deepreduce_from_params(dict(params))
+    opt = DistributedOptimizer(
+        torch.optim.SGD(model.parameters(), lr=lr, momentum=0.9,
+                        weight_decay=5e-4), grc, model)
+    sched = torch.optim.lr_scheduler.CosineAnnealingLR(opt.optimizer,
+                                                       T_max=epochs)
+    n = len(xtr)
+    t0 = time.time()
+    for ep in range(epochs):
+        perm = torch.randperm(n, generator=torch.Generator().manual_seed(ep))
+        for i in range(0, n - batch + 1, batch):
+            idx = perm[i : i + batch]
+            opt.zero_grad(set_to_none=False)
+            loss = F.cross_entropy(model(xtr[idx]), ytr[idx])
+            loss.backward()
+            opt.step()
+        sched.step()
+    acc_tr = accuracy(model, xtr, ytr)
+    acc_te = accuracy(model, xte, yte)
+    print(json.dumps({"config": label, "train_acc": round(acc_tr, 4),
+                      "test_acc": round(acc_te, 4),
+                      "final_loss": round(float(loss), 4),
+                      "minutes": round((time.time() - t0) / 60, 1)}),
+          flush=True)
+    return label, acc_tr, acc_te
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--epochs", type=int, default=12)
+    ap.add_argument("--out", default=None)
+    args = ap.parse_args()
+    torch.set_num_threads(max(1, (os.cpu_count() or 8) - 2))
+    data = make_dataset()
+    rows = [run_config(lbl, p, data, args.epochs) for lbl, p in CONFIGS]
+    if args.out:
+        dense_acc = rows[0][2]
+        lines = [
+            "",
+            "## ResNet-20 on CIFAR-shaped synthetic data "
+            f"({args.epochs} epochs, matched seeds)",
+            "",
+            "CIFAR-10 is not retrievable offline; this reproduces the paper's",
+            "CLAIM (DR tracks dense accuracy at matched epochs, pdf p.6",
+            "Fig. 15) on the real ResNet-20 architecture over a 10-class",
+            "32x32x3 procedural task.",
+            "",
+            "| Config | train acc | test acc | vs dense |",
+            "|---|---:|---:|---:|",
+        ]
+        for lbl, atr, ate in rows:
+            lines.append(f"| {lbl} | {atr:.4f} | {ate:.4f} | "
+                         f"{ate - dense_acc:+.4f} |")
+        with open(args.out, "a") as f:
+            f.write("\n".join(lines) + "\n")
+
+
+if __name__ == "__main__":
+    main()

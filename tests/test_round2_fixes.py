@@ -306,3 +306,22 @@ class TestBlockPFor:
         got = torch.zeros_like(t).scatter_(0, i2, v2)
         want = torch.zeros_like(t).scatter_(0, idx, vals)
         assert torch.allclose(got, want)
+
+
+def test_tensor_wire_bytes_measured_not_estimated():
+    """VERDICT weak item 7: after one fused exchange, the stats logger
+    must see the MEASURED per-tensor payload size for non-pipelined codecs
+    (rle/gzip/huffman), not the 8k fp32+int32 estimate."""
+    from deepreduce_amd.factory import grace_from_params
+
+    grc = grace_from_params({
+        "compressor": "topk", "memory": "residual",
+        "communicator": "allgather", "compress_ratio": 0.02,
+        "deepreduce": "index", "index": "rle",
+    })
+    torch.manual_seed(1)
+    named = [("w", torch.randn(50_000)), ("v", torch.randn(20_000))]
+    grc.step_many([(n, t.clone()) for n, t in named])
+    total = grc.last_wire_bytes
+    per_tensor = sum(grc._tensor_wire_bytes(n, t) for n, t in named)
+    assert per_tensor == total, (per_tensor, total)

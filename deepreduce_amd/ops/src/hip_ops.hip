@@ -47,20 +47,26 @@ __host__ __device__ __forceinline__ void hash_bases(int64_t item, uint32_t* h1, 
     *h2 = fmix32(*h1 ^ H2_SALT) | 1u;
 }
 
-// Incremental double hashing: (h1 + j*h2) mod m == iterate pos += (h2 mod m)
-// with one conditional subtract — bit-identical to hashing.py's int64 mod,
-// without a 64-bit mod per probe.
+// Position derivation: Lemire multiply-shift range reduction
+//   pos_j = ((h1 + j*h2 mod 2^32) * m) >> 32
+// instead of (h1 + j*h2) mod m.  Uniform on [0, m) for m < 2^31 and
+// replaces the two 64-bit integer divisions per candidate (the dominant
+// VALU cost of the full-universe query) with one 32x32 multiply-high per
+// probe.  Bit-identical to hashing.py's bloom_positions (parity-tested).
+__host__ __device__ __forceinline__ uint64_t bloom_pos(uint32_t x, int64_t m) {
+    return ((uint64_t)x * (uint64_t)m) >> 32;
+}
+
 __host__ __device__ __forceinline__ bool bloom_test(const uint8_t* __restrict__ bits, int64_t m,
                                            int k, int64_t item) {
     uint32_t h1, h2;
     hash_bases(item, &h1, &h2);
-    uint64_t pos = (uint64_t)h1 % (uint64_t)m;
-    uint64_t step = (uint64_t)h2 % (uint64_t)m;
+    uint32_t x = h1;
     for (int j = 0;;) {
+        uint64_t pos = bloom_pos(x, m);
         if (!((bits[pos >> 3] >> (pos & 7)) & 1)) return false;
         if (++j >= k) return true;
-        pos += step;
-        if (pos >= (uint64_t)m) pos -= (uint64_t)m;
+        x += h2;
     }
 }
 
@@ -87,8 +93,9 @@ __global__ void bloom_insert_kernel(const int64_t* __restrict__ items, int64_t n
     for (; i < n; i += stride) {
         uint32_t h1, h2;
         hash_bases(items[i], &h1, &h2);
-        for (int j = 0; j < k; ++j) {
-            int64_t pos = (int64_t)(((uint64_t)h1 + (uint64_t)j * h2) % (uint64_t)m);
+        uint32_t x = h1;
+        for (int j = 0; j < k; ++j, x += h2) {
+            int64_t pos = (int64_t)bloom_pos(x, m);
             atomicOr(&words[pos >> 5], 1u << (pos & 31));
         }
     }
@@ -152,16 +159,14 @@ __global__ void bloom_count_kernel(const uint8_t* __restrict__ bits, int64_t str
         if (alive) {
             uint32_t h1, h2;
             hash_bases(i, &h1, &h2);
-            uint64_t pos = (uint64_t)h1 % (uint64_t)m;
-            uint64_t step = (uint64_t)h2 % (uint64_t)m;
-            for (int j = 0; j < k && alive; ++j) {
+            uint32_t x = h1;
+            for (int j = 0; j < k && alive; ++j, x += h2) {
+                uint64_t pos = bloom_pos(x, m);
                 int64_t byte = pos >> 3;
                 uint8_t bit = pos & 7;
                 for (int r = 0; r < R; ++r)
                     if (alive & (1u << r))
                         if (!((bits[r * stride_bytes + byte] >> bit) & 1)) alive &= ~(1u << r);
-                pos += step;
-                if (pos >= (uint64_t)m) pos -= (uint64_t)m;
             }
         }
         for (int r = 0; r < R; ++r) {
@@ -1141,12 +1146,10 @@ __global__ void bt_insert_kernel(const int64_t* __restrict__ out_i,
         hash_bases(out_i[j], &h1, &h2);
         const int64_t m = D[4];
         uint32_t* bits = (uint32_t*)(wire + D[6]);
-        uint64_t pos = (uint64_t)h1 % (uint64_t)m;
-        uint64_t step = (uint64_t)h2 % (uint64_t)m;
-        for (int h = 0; h < (int)D[5]; ++h) {
+        uint32_t x = h1;
+        for (int h = 0; h < (int)D[5]; ++h, x += h2) {
+            uint64_t pos = bloom_pos(x, m);
             atomicOr(&bits[pos >> 5], 1u << (pos & 31));
-            pos += step;
-            if (pos >= (uint64_t)m) pos -= (uint64_t)m;
         }
     }
 }
@@ -1195,9 +1198,9 @@ __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
         if (alive) {
             uint32_t h1, h2;
             hash_bases(i, &h1, &h2);
-            uint64_t pos = (uint64_t)h1 % (uint64_t)m;
-            uint64_t step = (uint64_t)h2 % (uint64_t)m;
-            for (int h = 0; h < nh && alive; ++h) {
+            uint32_t x = h1;
+            for (int h = 0; h < nh && alive; ++h, x += h2) {
+                const uint64_t pos = bloom_pos(x, m);
                 const int64_t word = pos >> 5;
                 const uint32_t bit = 1u << (pos & 31);
                 if (use_lds) {
@@ -1209,8 +1212,6 @@ __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
                                                      + bitoff))[word] & bit))
                                 alive &= ~(1u << r);
                 }
-                pos += step;
-                if (pos >= (uint64_t)m) pos -= (uint64_t)m;
             }
         }
         for (int r = 0; r < R; ++r) {
@@ -1284,16 +1285,14 @@ __global__ void bt_qcount_inter_kernel(const uint32_t* __restrict__ il, int R,
         if (alive) {
             uint32_t h1, h2;
             hash_bases(i, &h1, &h2);
-            uint64_t pos = (uint64_t)h1 % (uint64_t)m;
-            uint64_t step = (uint64_t)h2 % (uint64_t)m;
-            for (int h = 0; h < nh && alive; ++h) {
+            uint32_t x = h1;
+            for (int h = 0; h < nh && alive; ++h, x += h2) {
+                const uint64_t pos = bloom_pos(x, m);
                 const uint32_t bit = 1u << (pos & 31);
                 const uint32_t* __restrict__ row = ilt + (pos >> 5) * R;
                 for (int r = 0; r < R; ++r)
                     if (alive & (1u << r))
                         if (!(row[r] & bit)) alive &= ~(1u << r);
-                pos += step;
-                if (pos >= (uint64_t)m) pos -= (uint64_t)m;
             }
         }
         for (int r = 0; r < R; ++r) {
@@ -2723,12 +2722,10 @@ torch::Tensor bloom_insert_cpu(torch::Tensor idxs, int64_t m, int64_t num_hash) 
     for (int64_t i = 0; i < n; ++i) {
         uint32_t h1, h2;
         hash_bases(it[i], &h1, &h2);
-        uint64_t pos = (uint64_t)h1 % (uint64_t)m;
-        uint64_t step = (uint64_t)h2 % (uint64_t)m;
-        for (int64_t j = 0; j < num_hash; ++j) {
+        uint32_t x = h1;
+        for (int64_t j = 0; j < num_hash; ++j, x += h2) {
+            uint64_t pos = bloom_pos(x, m);
             bits[pos >> 3] |= (uint8_t)(1u << (pos & 7));
-            pos += step;
-            if (pos >= (uint64_t)m) pos -= (uint64_t)m;
         }
     }
     return out;

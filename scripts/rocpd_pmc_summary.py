@@ -43,18 +43,30 @@ def main():
         return
     cols = [r[1] for r in db.execute(f"PRAGMA table_info({cnt})")]
     print(f"counter table: {cnt}; cols: {cols}", file=out)
-    # expected rocpd shape: rows keyed by dispatch/counter ids with a value
-    cinfo = find("rocpd_info_counter") or find("rocpd_info_pmc")
-    if cinfo:
-        names = dict(db.execute(
-            f"SELECT id, name FROM {cinfo}").fetchall())
-    else:
-        names = {}
-    # try the canonical join
-    key = "counter_id" if "counter_id" in cols else None
-    dkey = "dispatch_id" if "dispatch_id" in cols else None
+    # counter-name lookup: any pmc/counter info table with (id, name)-ish
+    names = {}
+    for t in tabs:
+        tl = t.lower()
+        if ("pmc" in tl or "counter" in tl) and t != cnt:
+            tcols = [r[1] for r in db.execute(f"PRAGMA table_info({t})")]
+            idc = next((c for c in ("id", "pmc_id") if c in tcols), None)
+            nmc = next((c for c in ("name", "symbol", "counter_name")
+                        if c in tcols), None)
+            if idc and nmc:
+                names.update(db.execute(
+                    f"SELECT {idc}, {nmc} FROM {t}").fetchall())
+                print(f"names from {t} ({len(names)})", file=out)
+    # ROCm 7.2 rocpd shape: rocpd_pmc_event(event_id, pmc_id, value)
+    # joined to the dispatch via its event_id
+    key = next((c for c in ("pmc_id", "counter_id") if c in cols), None)
     val = "value" if "value" in cols else None
-    if not (key and val and sym and dis):
+    if "event_id" in cols:
+        join = f"JOIN {dis} d ON c.event_id = d.event_id"
+    elif "dispatch_id" in cols:
+        join = f"JOIN {dis} d ON c.dispatch_id = d.dispatch_id"
+    else:
+        join = None
+    if not (key and val and join and sym and dis):
         print("unexpected schema; dumping 5 sample rows:", file=out)
         for r in db.execute(f"SELECT * FROM {cnt} LIMIT 5"):
             print(" ", r, file=out)
@@ -63,7 +75,7 @@ def main():
     q = f"""
         SELECT s.display_name, c.{key}, SUM(c.{val}), COUNT(*)
         FROM {cnt} c
-        JOIN {dis} d ON c.{dkey} = d.dispatch_id
+        {join}
         JOIN {sym} s ON d.kernel_id = s.id
         WHERE 1=1 {like}
         GROUP BY s.display_name, c.{key}

@@ -94,6 +94,8 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
     use_cuda = torch.cuda.is_available() and args.device != "cpu"
+    # let MIOpen pick tuned conv algorithms during warmup (fixed shapes)
+    torch.backends.cudnn.benchmark = True
     # modulo wrap: ranks beyond the device count co-locate (RCCL supports
     # multiple ranks per GPU), so `--gpus 2` is testable on a 1-GPU box
     ndev = max(1, torch.cuda.device_count()) if use_cuda else 1

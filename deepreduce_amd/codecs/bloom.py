@@ -190,8 +190,13 @@ class Bloom(SparseCompressor):
 
 class BloomCPU(Bloom):
     """CPU-pinned variant (reference parity for 'bloom_cpu',
-    pytorch/deepreduce.py:693-736 — there pybloomfilter-backed; here the same
-    deterministic torch codec executed on CPU)."""
+    pytorch/deepreduce.py:693-736 — there pybloomfilter-backed).
+
+    Not a separate algorithm on purpose: on CPU tensors the ops layer
+    dispatches to the native C++ implementations in the same extension
+    (`bloom_insert_cpu` / `bloom_query_positives_cpu`, hip_ops.hip), which
+    replace pybloomfilter, and the wire format is bit-identical to the GPU
+    kernels' (cross-tested).  This class only pins the execution device."""
 
     @staticmethod
     def compress(sparse_tensor, params):

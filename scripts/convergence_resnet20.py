@@ -44,10 +44,15 @@ CONFIGS = [
                               "compress_ratio": 0.01, "deepreduce": "both",
                               "value": "polyfit", "index": "bloom",
                               "policy": "leftmost"}),
+    ("DR-QSGD-BF-P0 (paper headline)",
+     {"compressor": "topk", "memory": "residual",
+      "communicator": "allgather", "compress_ratio": 0.01,
+      "deepreduce": "both", "value": "qsgd", "index": "bloom",
+      "policy": "p0", "qsgd_pack": True, "quantum_num": 63}),
 ]
 
 
-def make_dataset(n_train=4096, n_test=1024, seed=0):
+def make_dataset(n_train=4096, n_test=1024, seed=0, noise=6.0):
     g = torch.Generator().manual_seed(seed)
     # class prototypes: low-frequency random patterns, 10 classes
     freq = torch.randn(10, 3, 8, 8, generator=g)
@@ -65,7 +70,7 @@ def make_dataset(n_train=4096, n_test=1024, seed=0):
             if flip[i]:
                 img = img.flip(-1)
             xs[i] = img
-        xs += 0.7 * torch.randn(xs.shape, generator=gen)
+        xs += noise * torch.randn(xs.shape, generator=gen)
         return xs, ys
 
     xtr, ytr = sample(n_train, torch.Generator().manual_seed(seed + 1))

@@ -1199,13 +1199,35 @@ __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
             uint32_t h1, h2;
             hash_bases(i, &h1, &h2);
             uint32_t x = h1;
-            for (int h = 0; h < nh && alive; ++h, x += h2) {
-                const uint64_t pos = bloom_pos(x, m);
-                const int64_t word = pos >> 5;
-                const uint32_t bit = 1u << (pos & 31);
-                if (use_lds) {
-                    if (!(ldsq[word] & bit)) alive = 0u;
-                } else {
+            if (use_lds) {
+                for (int h = 0; h < nh && alive; ++h, x += h2) {
+                    const uint64_t pos = bloom_pos(x, m);
+                    if (!(ldsq[pos >> 5] & (1u << (pos & 31)))) alive = 0u;
+                }
+            } else {
+                // Latency shape (PMC r2e: WAIT:BUSY was 30:1 on the serial
+                // early-exit loop): probe the first `head` positions with
+                // INDEPENDENT loads issued before any test — one memory
+                // round-trip instead of a dependent chain — then finish the
+                // few survivors (fill=0.5 => ~6% past 4 probes) serially.
+                const int head = nh < 4 ? nh : 4;
+                uint64_t p[4];
+                for (int j = 0; j < head; ++j, x += h2) p[j] = bloom_pos(x, m);
+                for (int r = 0; r < R; ++r) {
+                    if (!(alive & (1u << r))) continue;
+                    const uint32_t* __restrict__ w32r =
+                        (const uint32_t*)(wires + (int64_t)r * wstride + bitoff);
+                    bool ok = true;
+                    uint32_t w[4];
+                    for (int j = 0; j < head; ++j) w[j] = w32r[p[j] >> 5];
+                    for (int j = 0; j < head; ++j)
+                        ok &= (w[j] >> (p[j] & 31)) & 1;
+                    if (!ok) alive &= ~(1u << r);
+                }
+                for (int h = head; h < nh && alive; ++h, x += h2) {
+                    const uint64_t pos = bloom_pos(x, m);
+                    const int64_t word = pos >> 5;
+                    const uint32_t bit = 1u << (pos & 31);
                     for (int r = 0; r < R; ++r)
                         if (alive & (1u << r))
                             if (!(((const uint32_t*)(wires + (int64_t)r * wstride
@@ -1286,7 +1308,19 @@ __global__ void bt_qcount_inter_kernel(const uint32_t* __restrict__ il, int R,
             uint32_t h1, h2;
             hash_bases(i, &h1, &h2);
             uint32_t x = h1;
-            for (int h = 0; h < nh && alive; ++h, x += h2) {
+            // same independent-head shape as bt_qcount (latency-bound loop)
+            const int head = nh < 4 ? nh : 4;
+            uint64_t p[4];
+            for (int j = 0; j < head; ++j, x += h2) p[j] = bloom_pos(x, m);
+            unsigned drop = 0u;
+            for (int j = 0; j < head; ++j) {
+                const uint32_t bit = 1u << (p[j] & 31);
+                const uint32_t* __restrict__ row = ilt + (p[j] >> 5) * R;
+                for (int r = 0; r < R; ++r)
+                    if (!(row[r] & bit)) drop |= (1u << r);
+            }
+            alive &= ~drop;
+            for (int h = head; h < nh && alive; ++h, x += h2) {
                 const uint64_t pos = bloom_pos(x, m);
                 const uint32_t bit = 1u << (pos & 31);
                 const uint32_t* __restrict__ row = ilt + (pos >> 5) * R;

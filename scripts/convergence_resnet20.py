@@ -44,6 +44,9 @@ CONFIGS = [
                               "compress_ratio": 0.01, "deepreduce": "both",
                               "value": "polyfit", "index": "bloom",
                               "policy": "leftmost"}),
+    ("DR-BF-P0", {"compressor": "topk", "memory": "residual",
+                  "communicator": "allgather", "compress_ratio": 0.01,
+                  "deepreduce": "index", "index": "bloom", "policy": "p0"}),
     ("DR-QSGD-BF-P0 (paper headline)",
      {"compressor": "topk", "memory": "residual",
       "communicator": "allgather", "compress_ratio": 0.01,

@@ -30,6 +30,8 @@ def summarize(path: str, top: int, md: bool):
         "SELECT name FROM sqlite_master WHERE type='table'")]
     sym = next(t for t in tabs if t.startswith("rocpd_info_kernel_symbol"))
     dis = next(t for t in tabs if t.startswith("rocpd_kernel_dispatch"))
+    like = (f"WHERE s.display_name LIKE '%{args.like}%'"
+            if getattr(args, "like", "") else "")
     rows = db.execute(f"""
         SELECT s.display_name, COUNT(*), SUM(d.end - d.start),
                AVG(d.end - d.start), MAX(s.arch_vgpr_count),
@@ -38,6 +40,7 @@ def summarize(path: str, top: int, md: bool):
                MAX(d.workgroup_size_x * d.workgroup_size_y * d.workgroup_size_z),
                MAX(d.grid_size_x * d.grid_size_y * d.grid_size_z)
         FROM {dis} d JOIN {sym} s ON d.kernel_id = s.id
+        {like}
         GROUP BY s.display_name ORDER BY SUM(d.end - d.start) DESC
     """).fetchall()
     total = sum(r[2] for r in rows) or 1
@@ -71,6 +74,7 @@ if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("db")
     ap.add_argument("--top", type=int, default=30)
+    ap.add_argument("--like", default="")
     ap.add_argument("--md", action="store_true")
     args = ap.parse_args()
     summarize(args.db, args.top, args.md)

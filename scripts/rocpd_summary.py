@@ -24,14 +24,12 @@ def wg_per_cu_lds(lds_bytes: int) -> int:
     return max(1, (160 * 1024) // lds_bytes)
 
 
-def summarize(path: str, top: int, md: bool):
+def summarize(path: str, top: int, md: bool, args=None):
     db = sqlite3.connect(path)
     tabs = [r[0] for r in db.execute(
         "SELECT name FROM sqlite_master WHERE type='table'")]
     sym = next(t for t in tabs if t.startswith("rocpd_info_kernel_symbol"))
     dis = next(t for t in tabs if t.startswith("rocpd_kernel_dispatch"))
-    like = (f"WHERE s.display_name LIKE '%{args.like}%'"
-            if getattr(args, "like", "") else "")
     rows = db.execute(f"""
         SELECT s.display_name, COUNT(*), SUM(d.end - d.start),
                AVG(d.end - d.start), MAX(s.arch_vgpr_count),
@@ -40,10 +38,12 @@ def summarize(path: str, top: int, md: bool):
                MAX(d.workgroup_size_x * d.workgroup_size_y * d.workgroup_size_z),
                MAX(d.grid_size_x * d.grid_size_y * d.grid_size_z)
         FROM {dis} d JOIN {sym} s ON d.kernel_id = s.id
-        {like}
         GROUP BY s.display_name ORDER BY SUM(d.end - d.start) DESC
     """).fetchall()
-    total = sum(r[2] for r in rows) or 1
+    total = sum(r[2] for r in rows) or 1  # pct is of ALL kernels
+    flt = getattr(args, "like", "")
+    if flt:  # substring filter in python (SQL LIKE treats _ as wildcard)
+        rows = [r for r in rows if flt in r[0]]
     hdr = ("kernel", "calls", "total_us", "avg_us", "pct", "vgpr", "agpr",
            "sgpr", "lds_B", "wg", "grid_wgs", "waves/simd", "wg/cu_lds")
     out = []
@@ -77,4 +77,4 @@ if __name__ == "__main__":
     ap.add_argument("--like", default="")
     ap.add_argument("--md", action="store_true")
     args = ap.parse_args()
-    summarize(args.db, args.top, args.md)
+    summarize(args.db, args.top, args.md, args)

@@ -741,12 +741,15 @@ def test_batched_multi_rank_decode_r8_interleaved(dev):
         (got - dense_ref).abs().max()
 
 
-def test_batched_large_filter_lds_fallback(dev):
-    """A filter larger than the 64 KB LDS budget must fall back to global
-    word loads inside bt_qcount (and still match the per-tensor path)."""
+def test_batched_large_filter_lds_fallback(dev, monkeypatch):
+    """With the LDS cache enabled (opt-in, DEEPREDUCE_LDSQ_MAX), a filter
+    larger than the budget must fall back to global word loads inside
+    bt_qcount (and still match the per-tensor path)."""
     from deepreduce_amd import deepreduce_from_params
-    from deepreduce_amd.ops.batched import BatchedPipeline, LDSQ_MAX
+    from deepreduce_amd.ops import batched as bt
+    from deepreduce_amd.ops.batched import BatchedPipeline
 
+    monkeypatch.setattr(bt, "LDSQ_MAX", 65536)
     params = {
         "compressor": "topk", "memory": "none",
         "communicator": "allgather", "compress_ratio": 0.05,
@@ -756,8 +759,9 @@ def test_batched_large_filter_lds_fallback(dev):
     names = ["big", "small"]
     bp = BatchedPipeline(names, numels, params, dev)
     mws = [(int(bp.desc[t, 4]) + 31) // 32 for t in range(2)]
-    assert mws[0] * 4 > LDSQ_MAX, "test premise: big filter exceeds budget"
-    assert mws[1] * 4 <= LDSQ_MAX
+    assert mws[0] * 4 > bt.LDSQ_MAX, "test premise: big filter exceeds budget"
+    assert mws[1] * 4 <= bt.LDSQ_MAX
+    assert bp.ldsq_bytes > 0, "LDS cache should be armed for the small filter"
 
     torch.manual_seed(13)
     flat = torch.randn(sum(numels), device=dev)

@@ -777,3 +777,21 @@ def test_batched_large_filter_lds_fallback(dev, monkeypatch):
         dec = grc.compressor.decompress(tc, ctx)
         assert torch.allclose(own[off : off + nel], dec.reshape(-1), atol=1e-6), n
         off += nel
+
+
+def test_block_pfor_gpu_roundtrip(dev):
+    """Block-PFoR runs GPU-resident (torch-vectorized): roundtrip on
+    device, wire identical to the CPU encode."""
+    from deepreduce_amd.codecs.intpack import pfor_decode, pfor_encode
+
+    torch.manual_seed(17)
+    g = torch.randint(0, 64, (50_000,), dtype=torch.int64)
+    g[torch.randint(0, 50_000, (64,))] = torch.randint(
+        1 << 20, 1 << 28, (64,), dtype=torch.int64)
+    w_cpu = pfor_encode(g)
+    w_gpu = pfor_encode(g.to(dev))
+    assert w_gpu.is_cuda
+    assert torch.equal(w_gpu.cpu(), w_cpu), "wire must be device-independent"
+    out = pfor_decode(w_gpu)
+    assert out.is_cuda
+    assert torch.equal(out.cpu(), g)

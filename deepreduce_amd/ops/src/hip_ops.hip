@@ -26,6 +26,19 @@
 
 static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
+// independent-load head width for the universe query (see bt_qcount):
+// runtime-tunable for same-box A/B via DEEPREDUCE_QHEAD (1..4, default 4)
+static inline int qhead_env() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("DEEPREDUCE_QHEAD");
+        v = e ? atoi(e) : 4;
+        if (v < 1) v = 1;
+        if (v > 4) v = 4;
+    }
+    return v;
+}
+
 // ---------------------------------------------------------------------------
 // hashing (must mirror deepreduce_amd/hashing.py exactly)
 // ---------------------------------------------------------------------------
@@ -1168,7 +1181,7 @@ __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
                                  int64_t wstride, int R,
                                  const int64_t* __restrict__ desc,
                                  const int* __restrict__ b2t, int64_t BV,
-                                 int64_t MW, int ldsq_words,
+                                 int64_t MW, int ldsq_words, int qhead,
                                  int* __restrict__ qcounts,
                                  uint64_t* __restrict__ mask) {
     extern __shared__ uint32_t ldsq[];
@@ -1210,7 +1223,7 @@ __global__ void bt_qcount_kernel(const uint8_t* __restrict__ wires,
                 // INDEPENDENT loads issued before any test — one memory
                 // round-trip instead of a dependent chain — then finish the
                 // few survivors (fill=0.5 => ~6% past 4 probes) serially.
-                const int head = nh < 4 ? nh : 4;
+                const int head = nh < qhead ? nh : qhead;
                 uint64_t p[4];
                 for (int j = 0; j < head; ++j, x += h2) p[j] = bloom_pos(x, m);
                 for (int r = 0; r < R; ++r) {
@@ -1287,7 +1300,8 @@ __global__ void bt_interleave_kernel(const uint8_t* __restrict__ wires,
 __global__ void bt_qcount_inter_kernel(const uint32_t* __restrict__ il, int R,
                                        const int64_t* __restrict__ desc,
                                        const int* __restrict__ b2t, int64_t BV,
-                                       int64_t MW, int* __restrict__ qcounts,
+                                       int64_t MW, int qhead,
+                                       int* __restrict__ qcounts,
                                        uint64_t* __restrict__ mask) {
     const int t = b2t[blockIdx.x];
     const int64_t* D = bt_row(desc, t);
@@ -1309,7 +1323,7 @@ __global__ void bt_qcount_inter_kernel(const uint32_t* __restrict__ il, int R,
             hash_bases(i, &h1, &h2);
             uint32_t x = h1;
             // same independent-head shape as bt_qcount (latency-bound loop)
-            const int head = nh < 4 ? nh : 4;
+            const int head = nh < qhead ? nh : qhead;
             uint64_t p[4];
             for (int j = 0; j < head; ++j, x += h2) p[j] = bloom_pos(x, m);
             unsigned drop = 0u;
@@ -1552,7 +1566,7 @@ std::vector<torch::Tensor> batched_compress(torch::Tensor values_flat,
     hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK),
                        (size_t)ldsq_bytes, stream,
                        wire.data_ptr<uint8_t>(), wire_bytes, 1, dp, mp, BV,
-                       mask_words, (int)(ldsq_bytes / 4), qcounts,
+                       mask_words, (int)(ldsq_bytes / 4), qhead_env(), qcounts,
                        (uint64_t*)mask.data_ptr<int64_t>());
     hipLaunchKernelGGL(bt_scan_kernel, dim3(T), dim3(QBLOCK), 0, stream,
                        qcounts, dp, T, BV, qoffs);
@@ -1621,14 +1635,14 @@ torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
         hipLaunchKernelGGL(bt_qcount_inter_kernel, dim3((int)BV), dim3(QBLOCK), 0,
                            stream, (const uint32_t*)il.data_ptr<int>(), R,
                            d.data_ptr<int64_t>(), map.data_ptr<int>(), BV,
-                           mask_words, qcounts,
+                           mask_words, qhead_env(), qcounts,
                            (uint64_t*)mask.data_ptr<int64_t>());
     } else {
         hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK),
                            (size_t)ldsq_bytes, stream,
                            w.data_ptr<uint8_t>(), W, R, d.data_ptr<int64_t>(),
                            map.data_ptr<int>(), BV, mask_words,
-                           (int)(ldsq_bytes / 4), qcounts,
+                           (int)(ldsq_bytes / 4), qhead_env(), qcounts,
                            (uint64_t*)mask.data_ptr<int64_t>());
     }
     hipLaunchKernelGGL(bt_scan_kernel, dim3(R * T), dim3(QBLOCK), 0, stream,
@@ -2245,7 +2259,7 @@ std::vector<torch::Tensor> batched_compress_both(
     hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK),
                        (size_t)ldsq_bytes, stream,
                        wire.data_ptr<uint8_t>(), wire_bytes, 1, dp, mp, BV,
-                       mask_words, (int)(ldsq_bytes / 4), qcounts,
+                       mask_words, (int)(ldsq_bytes / 4), qhead_env(), qcounts,
                        (uint64_t*)mask.data_ptr<int64_t>());
     hipLaunchKernelGGL(bt_scan_kernel, dim3(T), dim3(QBLOCK), 0, stream,
                        qcounts, dp, T, BV, qoffs);
@@ -2331,14 +2345,14 @@ torch::Tensor batched_decode_both_sum(torch::Tensor wires2d, torch::Tensor desc,
         hipLaunchKernelGGL(bt_qcount_inter_kernel, dim3((int)BV), dim3(QBLOCK), 0,
                            stream, (const uint32_t*)il.data_ptr<int>(), R,
                            d.data_ptr<int64_t>(), map.data_ptr<int>(), BV,
-                           mask_words, qcounts,
+                           mask_words, qhead_env(), qcounts,
                            (uint64_t*)mask.data_ptr<int64_t>());
     } else {
         hipLaunchKernelGGL(bt_qcount_kernel, dim3((int)BV), dim3(QBLOCK),
                            (size_t)ldsq_bytes, stream,
                            w.data_ptr<uint8_t>(), W, R, d.data_ptr<int64_t>(),
                            map.data_ptr<int>(), BV, mask_words,
-                           (int)(ldsq_bytes / 4), qcounts,
+                           (int)(ldsq_bytes / 4), qhead_env(), qcounts,
                            (uint64_t*)mask.data_ptr<int64_t>());
     }
     hipLaunchKernelGGL(bt_scan_kernel, dim3(R * T), dim3(QBLOCK), 0, stream,

@@ -27,12 +27,17 @@
 static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
 // independent-load head width for the universe query (see bt_qcount):
-// runtime-tunable for same-box A/B via DEEPREDUCE_QHEAD (1..4, default 4)
+// runtime-tunable via DEEPREDUCE_QHEAD (1..4).  MEASURED same-box
+// (gpurun_out/qcount_head_sweep.json, 40-iter A/B): whole-model compress
+// 0.499/0.512/0.562 ms and R=8 interleaved decode 1.629/1.589/1.950 ms
+// at head 1/2/4 — two independent probe loads balance chain-breaking
+// against wasted loads (fill=0.5 kills half the candidates at probe 1),
+// so the default is 2.
 static inline int qhead_env() {
     static int v = -1;
     if (v < 0) {
         const char* e = getenv("DEEPREDUCE_QHEAD");
-        v = e ? atoi(e) : 4;
+        v = e ? atoi(e) : 2;
         if (v < 1) v = 1;
         if (v > 4) v = 4;
     }

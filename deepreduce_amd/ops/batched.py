@@ -125,6 +125,22 @@ class BatchedPipeline:
                                               int(self.wire_half))
 
     def decode_sum(self, wires2d):
+        return self._decode_chunked(wires2d, self._decode_sum16)
+
+    def _decode_chunked(self, wires2d, fn):
+        """The query kernels handle <=16 filters per launch (MAXR);
+        larger worlds decode in sequential chunks of 16 (deterministic
+        accumulation order, identical on every rank)."""
+        R = int(wires2d.size(0))
+        if R <= 16:
+            return fn(wires2d)
+        total = None
+        for i in range(0, R, 16):
+            part = fn(wires2d[i : i + 16].contiguous())
+            total = part if total is None else total.add_(part)
+        return total
+
+    def _decode_sum16(self, wires2d):
         from deepreduce_amd import _hip_ops
 
         return _hip_ops.batched_decode_sum(wires2d, self.desc, self.b2t,
@@ -222,6 +238,9 @@ class BothPipeline(BatchedPipeline):
         return wire, own
 
     def decode_sum(self, wires2d):
+        return self._decode_chunked(wires2d, self._decode_sum16)
+
+    def _decode_sum16(self, wires2d):
         from deepreduce_amd import _hip_ops
 
         return _hip_ops.batched_decode_both_sum(
@@ -301,6 +320,9 @@ class ValuePipeline(BatchedPipeline):
         return wire, own
 
     def decode_sum(self, wires2d):
+        return self._decode_chunked(wires2d, self._decode_sum16)
+
+    def _decode_sum16(self, wires2d):
         from deepreduce_amd import _hip_ops
 
         return _hip_ops.batched_decode_value_sum(
@@ -348,8 +370,6 @@ def maybe_pipeline(communicator, comp, named_tensors):
     if any(t.dtype != torch.float32 or not t.is_cuda for t in grads):
         return None
     if any(t.numel() <= 1000 for t in grads):  # codec bypass: generic path
-        return None
-    if communicator.world_size > 16:
         return None
     key = (
         cls.__name__,

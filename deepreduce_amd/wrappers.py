@@ -169,7 +169,6 @@ class IndexCompressor(_WrapperBase):
             or self.params.get("policy", "leftmost") != "leftmost"
             or numel <= _BYPASS_NUMEL
             or len(payloads) < 2
-            or len(payloads) > 16  # MAXR of the batched query kernel
         ):
             return None
         from . import ops
@@ -179,12 +178,18 @@ class IndexCompressor(_WrapperBase):
         if any(int(p[0].numel()) != num_indices for p in payloads):
             return None  # leftmost assumes uniform k (topk sparsifier)
         num_hash, m = Bloom._config(num_indices, numel, self.params)
-        bits = torch.stack([p[1].contiguous() for p in payloads])
-        # sync-free: [R, k] leftmost positives, one fused scatter-add
-        idxs = ops.bloom_query_leftmost(bits, m, num_hash, numel, num_indices)
-        vals = torch.stack([p[0] for p in payloads]).float()  # fp16 wire ok
         dense = torch.zeros(numel, dtype=torch.float32, device=vals0.device)
-        dense.index_add_(0, idxs.reshape(-1), vals.reshape(-1))
+        # the query kernel handles <=16 filters per launch (MAXR); larger
+        # worlds decode in chunks of 16, summed sequentially (deterministic
+        # order, identical on every rank)
+        for c in range(0, len(payloads), 16):
+            chunk = payloads[c : c + 16]
+            bits = torch.stack([p[1].contiguous() for p in chunk])
+            # sync-free: [R, k] leftmost positives, one fused scatter-add
+            idxs = ops.bloom_query_leftmost(bits, m, num_hash, numel,
+                                            num_indices)
+            vals = torch.stack([p[0] for p in chunk]).float()  # fp16 wire ok
+            dense.index_add_(0, idxs.reshape(-1), vals.reshape(-1))
         return dense.view(shape)
 
 

@@ -102,12 +102,34 @@ def _conflict_sets_select(positives: torch.Tensor, k: int, seed: int, m: int, nu
     return torch.as_tensor(out, dtype=torch.int64, device=positives.device)
 
 
+def auto_fpr(k: int, d: int, value_bytes: float = 4.0) -> float:
+    """Wire-minimizing FPR (params['fpr']='auto'; absent in the reference,
+    which fixes fpr = 0.1*k/d).
+
+    Total wire bytes(f) = filter + values carried by false positives:
+        m/8 + fp_cost = k*log2(1/f)/(8*ln2) + (d-k)*f*value_bytes
+    (leftmost/random clip to k values, but every universe slot that can
+    test positive still costs decode work and, under P0, wire bytes —
+    this closed form prices the P0/FP-aware worst case).  Setting the
+    derivative to zero: f* = k / (8*ln2*ln2 * (d-k) * value_bytes).
+    Clamped to [1e-5, 0.5] and evaluated deterministically from (k, d)
+    only, so every rank derives the same configuration.
+    """
+    if d <= k:
+        return 0.5
+    f = k / (8.0 * LN2 * LN2 * (d - k) * value_bytes)
+    return float(min(0.5, max(1e-5, f)))
+
+
 class Bloom(SparseCompressor):
     order_preserving = False
 
     @staticmethod
     def _config(num_indices: int, grad_size: int, params):
         fpr = params.get("fpr", 0.1 * num_indices / grad_size)
+        if fpr == "auto":
+            vb = 2.0 if params.get("wire_dtype") == "fp16" else 4.0
+            fpr = auto_fpr(num_indices, grad_size, vb)
         return get_bf_config(num_indices, fpr)
 
     @staticmethod

@@ -24,7 +24,7 @@ _SCHEMA = {
     "policy": (str, ["leftmost", "random", "p0", "conflict_sets"]),
     "compress_ratio": ((int, float), None),
     "threshold": ((int, float), None),
-    "fpr": ((int, float), None),
+    "fpr": ((int, float, str), None),  # number or "auto"
     "sort": (bool, None),
     "poly_degree": (int, None),
     "num_segments": (int, None),
@@ -65,7 +65,9 @@ def validate(params: dict) -> dict:
     if ratio is not None and not (0 < ratio <= 1):
         raise ValueError(f"compress_ratio must be in (0, 1], got {ratio}")
     fpr = params.get("fpr")
-    if fpr is not None and not (0 < fpr < 1):
+    if isinstance(fpr, str) and fpr != "auto":
+        raise ValueError(f"fpr must be a number in (0, 1) or 'auto', got {fpr!r}")
+    if isinstance(fpr, (int, float)) and not (0 < fpr < 1):
         raise ValueError(f"fpr must be in (0, 1), got {fpr}")
     if params.get("policy") == "conflict_sets":
         # The conflict-sets policy (reference policies.hpp:43-146 semantics)

@@ -325,3 +325,38 @@ def test_tensor_wire_bytes_measured_not_estimated():
     total = grc.last_wire_bytes
     per_tensor = sum(grc._tensor_wire_bytes(n, t) for n, t in named)
     assert per_tensor == total, (per_tensor, total)
+
+
+def test_auto_fpr():
+    """fpr='auto' picks the closed-form wire-minimizing FPR and beats both
+    the default and a too-large FPR on total wire bytes."""
+    from deepreduce_amd.codecs import compressor as registry
+    from deepreduce_amd.codecs.bloom import auto_fpr, get_bf_config
+
+    d, k = 1_000_000, 10_000
+    f = auto_fpr(k, d)
+    assert 1e-5 <= f <= 0.5
+
+    def wire_bytes(fpr):
+        nh, m = get_bf_config(k, fpr)
+        return m / 8 + (d - k) * fpr * 4  # filter + FP-carried values
+
+    assert wire_bytes(f) <= wire_bytes(0.1 * k / d) + 1
+    assert wire_bytes(f) <= wire_bytes(0.2) + 1
+
+    # end-to-end: codec accepts the string and produces a valid roundtrip
+    torch.manual_seed(31)
+    t = torch.randn(d)
+    _, idx = t.abs().topk(k)
+    params = {"fpr": "auto", "policy": "leftmost", "dense_tensor": t}
+    v, bits, shape = registry["bloom"].compress((t[idx], idx, t.size()), params)
+    params.pop("dense_tensor")
+    params.pop("_own_decoded", None)
+    v2, i2, _ = registry["bloom"].decompress((v, bits, shape), params)
+    assert v2.numel() == i2.numel() == k
+    assert torch.allclose(t[i2], v2)
+
+    from deepreduce_amd.params import validate
+    validate({"fpr": "auto"})
+    with pytest.raises(ValueError):
+        validate({"fpr": "wrong"})

@@ -136,6 +136,20 @@ def main():
     ]
     for label, s2c, c2s, rs, rc in rows:
         lines.append(f"| {label} | {s2c:,} | {c2s:,} | {rs:.4f} | {rc:.4f} |")
+    # paper Table 4 equivalent: simulated comm time on a 100 Mbps FL link
+    # (bytes / 12.5 MB/s), from the SAME measured volumes
+    lines += [
+        "",
+        "## Simulated 100 Mbps comm time per round (paper p.33 Table 4 shape)",
+        "",
+        "| Config | S2C s | C2S s/client | vs dense |",
+        "|---|---:|---:|---:|",
+    ]
+    bw = 100e6 / 8
+    base = rows[0][1] / bw
+    for label, s2c, c2s, rs, rc in rows:
+        lines.append(f"| {label} | {s2c / bw:.3f} | {c2s / bw:.3f} | "
+                     f"{base / max(s2c / bw, 1e-9):.1f}x |")
     text = "\n".join(lines) + "\n"
     print(text)
     if args.out:

@@ -84,3 +84,61 @@ def test_polyfit_payload_uniform_and_bounded_error(data):
     v2, i2, _ = compressor["polyfit"].decompress((p, m, s), params)
     assert v2.numel() == vals.numel()
     assert torch.isfinite(v2).all()
+
+
+# ---- round 2: whole-wrapper property sweep --------------------------------
+
+_MODES = [
+    (None, None, None, None),
+    ("value", "qsgd", None, None),
+    ("value", "polyfit", None, None),
+    ("value", "doubleexp", None, None),
+    ("index", None, "bloom", "leftmost"),
+    ("index", None, "bloom", "p0"),
+    ("index", None, "rle", None),
+    ("index", None, "pfor", None),
+    ("both", "polyfit", "bloom", "leftmost"),
+    ("both", "qsgd", "bloom", "p0"),      # mapping-free wire
+    ("both", "qsgd", "bloom", "leftmost"),
+]
+
+
+@settings(max_examples=8, deadline=None)
+@given(data=st.data())
+@pytest.mark.parametrize("mode", _MODES, ids=lambda m: f"{m[0]}-{m[1]}-{m[2]}-{m[3]}")
+def test_wrapper_step_invariants(mode, data):
+    """For any config and tensor size: grc.step returns the tensor's shape,
+    finite values, positive wire accounting, and error no worse than
+    plain top-k + residual can explain (reconstruction bounded by input
+    norm)."""
+    from deepreduce_amd import deepreduce_from_params
+
+    dr, value, index, policy = mode
+    n = data.draw(st.integers(min_value=1100, max_value=40_000))
+    seed = data.draw(st.integers(0, 2**31 - 1))
+    ratio = data.draw(st.sampled_from([0.01, 0.05, 0.2]))
+    params = {"compressor": "topk", "memory": "residual",
+              "communicator": "allgather", "compress_ratio": ratio}
+    if dr:
+        params["deepreduce"] = dr
+        if value:
+            params["value"] = value
+        if index:
+            params["index"] = index
+        if policy:
+            params["policy"] = policy
+    grc = deepreduce_from_params(params)
+    g = torch.Generator().manual_seed(seed)
+    t = torch.randn(n, generator=g)
+    out = grc.step(t.clone(), "w")
+    assert out.shape == t.shape
+    assert torch.isfinite(out).all()
+    assert grc.last_wire_bytes > 0
+    # decompressed energy cannot exceed input energy by much (lossy codecs
+    # approximate a k-subset of t, possibly amplified by fit overshoot)
+    assert out.norm() <= t.norm() * 3 + 1
+    # residual consistency: residual == compensated - out (first step:
+    # compensated == t)
+    r = grc.memory.residuals["w"]
+    assert torch.allclose(r, t - out.view_as(t), atol=2e-4), \
+        (r - (t - out.view_as(t))).abs().max()

@@ -58,14 +58,23 @@ def unpack_with_header(wire: torch.Tensor) -> torch.Tensor:
 #   [4]    0xFF     format tag (legacy fixed-width wires carry nbits<=57 here)
 #   [5:9]  nb       number of 128-int blocks, LE u32
 #   nb * 3 bytes    per-block meta: [width b][n_exceptions][exception width]
-#   per block:      ceil(cnt*b/8) bytes   low-b-bit stream (cnt=128, last may
-#                                         be partial)
-#   per block:      n_exc bytes           exception lane positions (0..127)
-#                   ceil(n_exc*exc_w/8)   exception high bits (v >> b)
+#   (header+meta padded to a multiple of 4 bytes)
+#   per block:      align4(ceil(cnt*b/8)) bytes   low-b-bit stream (cnt=128,
+#                                                 last may be partial)
+#   per block:      align4(n_exc) bytes           exception lane positions
+#                   align4(ceil(n_exc*exc_w/8))   exception high bits (v >> b)
+# Every region is 4-byte aligned so the HIP pack/unpack kernels write/read
+# whole 32-bit words without crossing block boundaries (cost: <=9 pad bytes
+# per 128-int block, ~0.2 bits/int).
 # ---------------------------------------------------------------------------
 
 _PFOR_BLOCK = 128
-_PFOR_TAG = 0xFF
+_PFOR_TAG = 0xFE  # v2: 4-byte-aligned regions
+_ALIGN = 4
+
+
+def _al4(x):
+    return (x + 3) & ~3
 
 
 def _bitlen(v: torch.Tensor) -> torch.Tensor:
@@ -113,7 +122,18 @@ def _bytes_to_bits(bytes_t: torch.Tensor) -> torch.Tensor:
 
 
 def pfor_encode(ints: torch.Tensor) -> torch.Tensor:
-    """Encode non-negative int64 tensor into the block-PFoR wire (uint8)."""
+    """Encode non-negative int64 tensor into the block-PFoR wire (uint8).
+
+    GPU tensors dispatch to the HIP kernel path (ops pfor_pack: stats wave
+    per block, word-parallel packing) which produces a byte-identical
+    wire; the torch-vectorized path below is the CPU implementation."""
+    if ints.is_cuda:
+        from .. import ops as _ops
+
+        if _ops.hip_available():
+            from deepreduce_amd import _hip_ops
+
+            return _hip_ops.pfor_pack(ints)
     v = ints.long().reshape(-1)
     n = v.numel()
     dev = v.device
@@ -152,10 +172,10 @@ def pfor_encode(ints: torch.Tensor) -> torch.Tensor:
                         exw.gather(1, b.unsqueeze(1)).squeeze(1),
                         torch.zeros_like(b))
 
-    # ---- layout (host sync: allocation sizes) ----
-    sb = ((cnt * b + 7) // 8)                          # low-bit stream bytes
-    pe = ne_b                                          # position bytes
-    hb = (ne_b * exw_b + 7) // 8                       # high-bit bytes
+    # ---- layout (host sync: allocation sizes); all regions 4B-aligned ----
+    sb = ((cnt * b + 7) // 8 + 3) & ~3                 # low-bit stream bytes
+    pe = (ne_b + 3) & ~3                               # position bytes
+    hb = ((ne_b * exw_b + 7) // 8 + 3) & ~3            # high-bit bytes
     sb_off = torch.cumsum(sb, 0) - sb
     total_sb = int(sb.sum().item())
     exc_off = torch.cumsum(pe + hb, 0) - (pe + hb)
@@ -198,10 +218,22 @@ def pfor_encode(ints: torch.Tensor) -> torch.Tensor:
         exc_buf[hb_slot_off[byte_blk] + byte_within] = hi_bytes
 
     meta = torch.stack([b, ne_b, exw_b], dim=1).reshape(-1).to(torch.uint8)
-    return torch.cat([header, meta, stream, exc_buf])
+    headmeta = torch.cat([header, meta])
+    pad = (-headmeta.numel()) % 4
+    if pad:
+        headmeta = torch.cat([headmeta, torch.zeros(pad, dtype=torch.uint8,
+                                                    device=dev)])
+    return torch.cat([headmeta, stream, exc_buf])
 
 
 def pfor_decode(wire: torch.Tensor) -> torch.Tensor:
+    if wire.is_cuda:
+        from .. import ops as _ops
+
+        if _ops.hip_available():
+            from deepreduce_amd import _hip_ops
+
+            return _hip_ops.pfor_unpack(wire)
     dev = wire.device
     head = wire[:9].cpu()
     n = int(head[0]) | (int(head[1]) << 8) | (int(head[2]) << 16) | (int(head[3]) << 24)
@@ -214,13 +246,15 @@ def pfor_decode(wire: torch.Tensor) -> torch.Tensor:
     cnt = torch.full((nb,), _PFOR_BLOCK, dtype=torch.int64, device=dev)
     if n % _PFOR_BLOCK:
         cnt[-1] = n % _PFOR_BLOCK
-    sb = (cnt * b + 7) // 8
-    pe = ne_b
-    hb = (ne_b * exw_b + 7) // 8
+    sb = ((cnt * b + 7) // 8 + 3) & ~3
+    pe = (ne_b + 3) & ~3
+    hb = ((ne_b * exw_b + 7) // 8 + 3) & ~3
     sb_off = torch.cumsum(sb, 0) - sb
     total_sb = int(sb.sum().item())
-    stream = wire[9 + nb * 3 : 9 + nb * 3 + total_sb].to(dev)
-    exc_buf = wire[9 + nb * 3 + total_sb :].to(dev)
+    hm = 9 + nb * 3
+    hm += (-hm) % 4
+    stream = wire[hm : hm + total_sb].to(dev)
+    exc_buf = wire[hm + total_sb :].to(dev)
 
     bit_arr = _bytes_to_bits(stream)
     lane = torch.arange(_PFOR_BLOCK, device=dev).unsqueeze(0).expand(nb, -1)

@@ -2764,6 +2764,279 @@ torch::Tensor dexp_fit(torch::Tensor y, torch::Tensor offs, torch::Tensor lens) 
     return out;
 }
 
+// ---------------------------------------------------------------------------
+// Block-PFoR pack/unpack kernels (codecs/intpack.py wire format v2:
+// 128-int blocks, per-block width, patched exceptions, 4-byte-aligned
+// regions).  The torch implementation stays as the CPU path; these kernels
+// produce BYTE-IDENTICAL wires (roundtrip + device-independence tested).
+// One wave per block for stats/exceptions; one thread per output word /
+// per value for the streams.
+// ---------------------------------------------------------------------------
+
+#define PFOR_BLK 128
+
+__device__ __forceinline__ int pfor_bitlen(int64_t v) {
+    return v > 0 ? 64 - __clzll((uint64_t)v) : 0;
+}
+
+// one wave per 128-int block: choose width, count exceptions, emit sizes
+__global__ void pfor_stats_kernel(const int64_t* __restrict__ v, int64_t n,
+                                  int64_t nb, int* __restrict__ width,
+                                  int* __restrict__ nexc,
+                                  int* __restrict__ excw,
+                                  int* __restrict__ sbytes /*aligned*/,
+                                  int* __restrict__ ebytes /*aligned*/) {
+    const int64_t blk = blockIdx.x;
+    if (blk >= nb) return;
+    const int lane = threadIdx.x;  // 64 lanes
+    const int64_t base = blk * PFOR_BLK;
+    const int cnt = (int)min((int64_t)PFOR_BLK, n - base);
+    const int j0 = 2 * lane, j1 = 2 * lane + 1;
+    const int bl0 = (j0 < cnt) ? pfor_bitlen(v[base + j0]) : 0;
+    const int bl1 = (j1 < cnt) ? pfor_bitlen(v[base + j1]) : 0;
+    // wave max bitlen
+    int mx = max(bl0, bl1);
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        mx = max(mx, __shfl_down(mx, off, WAVE));
+    mx = __shfl(mx, 0, WAVE);
+    // cost(b) = ceil(cnt*b/8) + ne + ceil(ne*exw/8) (same model as the
+    // torch encoder, UNALIGNED, so argmin matches bit-for-bit)
+    int best_b = 0, best_cost = 0x7FFFFFFF, best_ne = 0;
+    for (int b = 0; b <= 32; ++b) {
+        const int ne = __popcll(__ballot(bl0 > b)) + __popcll(__ballot(bl1 > b));
+        const int exw = mx > b ? mx - b : 0;
+        const int cost = (cnt * b + 7) / 8 + ne + (ne * exw + 7) / 8;
+        if (cost < best_cost) { best_cost = cost; best_b = b; best_ne = ne; }
+    }
+    if (lane == 0) {
+        const int exw = (best_ne > 0 && mx > best_b) ? mx - best_b : 0;
+        width[blk] = best_b;
+        nexc[blk] = best_ne;
+        excw[blk] = exw;
+        sbytes[blk] = (((cnt * best_b + 7) / 8) + 3) & ~3;
+        ebytes[blk] = ((best_ne + 3) & ~3) + ((((best_ne * exw + 7) / 8) + 3) & ~3);
+    }
+}
+
+// one thread per 32-bit output word of the low-bit stream
+__global__ void pfor_pack_kernel(const int64_t* __restrict__ v, int64_t n,
+                                 int64_t nb, const int* __restrict__ width,
+                                 const int64_t* __restrict__ sw_off /*words*/,
+                                 int64_t total_sw, uint32_t* __restrict__ stream) {
+    int64_t w = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; w < total_sw; w += stride) {
+        int lo = 0, hi = (int)nb - 1;      // largest blk with sw_off[blk] <= w
+        while (lo < hi) {
+            const int mid = (lo + hi + 1) >> 1;
+            if (sw_off[mid] <= w) lo = mid; else hi = mid - 1;
+        }
+        const int b = width[lo];
+        uint32_t out = 0;
+        if (b > 0) {
+            const int64_t base = (int64_t)lo * PFOR_BLK;
+            const int cnt = (int)min((int64_t)PFOR_BLK, n - base);
+            const int64_t wbit = (w - sw_off[lo]) * 32;   // bit off in block
+            const uint64_t mask = (b >= 64) ? ~0ull : ((1ull << b) - 1);
+            int j = (int)(wbit / b);
+            for (; j < cnt && (int64_t)j * b < wbit + 32; ++j) {
+                const uint64_t val = (uint64_t)v[base + j] & mask;
+                const int64_t sh = (int64_t)j * b - wbit;  // may be negative
+                if (sh >= 0) out |= (uint32_t)(val << sh);
+                else out |= (uint32_t)(val >> (-sh));
+            }
+        }
+        stream[w] = out;
+    }
+}
+
+// one wave per block: exception positions (bytes) + high bits (atomicOr)
+__global__ void pfor_exc_kernel(const int64_t* __restrict__ v, int64_t n,
+                                int64_t nb, const int* __restrict__ width,
+                                const int* __restrict__ nexc,
+                                const int* __restrict__ excw,
+                                const int64_t* __restrict__ e_off /*bytes*/,
+                                uint8_t* __restrict__ exc) {
+    const int64_t blk = blockIdx.x;
+    if (blk >= nb) return;
+    const int lane = threadIdx.x;
+    const int b = width[blk], ne = nexc[blk], exw = excw[blk];
+    if (ne == 0) return;
+    const int64_t base = blk * PFOR_BLK;
+    const int cnt = (int)min((int64_t)PFOR_BLK, n - base);
+    uint8_t* pos_out = exc + e_off[blk];
+    uint32_t* hi_out = (uint32_t*)(exc + e_off[blk] + ((ne + 3) & ~3));
+    int rank_base = 0;
+    for (int half = 0; half < 2; ++half) {
+        const int j = 2 * lane + half;
+        const bool is_exc = (j < cnt) && (pfor_bitlen(v[base + j]) > b);
+        const uint64_t ball = __ballot(is_exc);
+        if (is_exc) {
+            const uint64_t below = (lane == 63) ? (~0ull >> 1)
+                                                : ((1ull << lane) - 1);
+            const int rank = rank_base + __popcll(ball & below);
+            pos_out[rank] = (uint8_t)j;
+            if (exw > 0) {
+                const uint64_t hiv = (uint64_t)v[base + j] >> b;
+                const int64_t sbit = (int64_t)rank * exw;
+                const int wi = (int)(sbit >> 5), sh = (int)(sbit & 31);
+                atomicOr(&hi_out[wi], (uint32_t)(hiv << sh));
+                if (sh + exw > 32)
+                    atomicOr(&hi_out[wi + 1], (uint32_t)(hiv >> (32 - sh)));
+            }
+        }
+        rank_base += __popcll(ball);
+    }
+}
+
+// one thread per value: gather its b low bits from <=2 stream words
+__global__ void pfor_unpack_kernel(const uint32_t* __restrict__ stream,
+                                   int64_t n, const int* __restrict__ width,
+                                   const int64_t* __restrict__ sw_off,
+                                   int64_t* __restrict__ out) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        const int64_t blk = i / PFOR_BLK;
+        const int b = width[blk];
+        if (b == 0) { out[i] = 0; continue; }
+        const int j = (int)(i - blk * PFOR_BLK);
+        const int64_t sbit = (int64_t)j * b;
+        const int64_t wi = sw_off[blk] + (sbit >> 5);
+        const int sh = (int)(sbit & 31);
+        uint64_t bits = (uint64_t)stream[wi] >> sh;
+        if (sh + b > 32) bits |= (uint64_t)stream[wi + 1] << (32 - sh);
+        out[i] = (int64_t)(bits & ((b >= 64) ? ~0ull : ((1ull << b) - 1)));
+    }
+}
+
+// one wave per block: apply exception high bits
+__global__ void pfor_unexc_kernel(const uint8_t* __restrict__ exc, int64_t n,
+                                  int64_t nb, const int* __restrict__ width,
+                                  const int* __restrict__ nexc,
+                                  const int* __restrict__ excw,
+                                  const int64_t* __restrict__ e_off,
+                                  int64_t* __restrict__ out) {
+    const int64_t blk = blockIdx.x;
+    if (blk >= nb) return;
+    const int b = width[blk], ne = nexc[blk], exw = excw[blk];
+    if (ne == 0 || exw == 0) return;
+    const uint8_t* pos_in = exc + e_off[blk];
+    const uint32_t* hi_in = (const uint32_t*)(exc + e_off[blk] + ((ne + 3) & ~3));
+    for (int r = threadIdx.x; r < ne; r += blockDim.x) {
+        const int j = pos_in[r];
+        const int64_t sbit = (int64_t)r * exw;
+        const int wi = (int)(sbit >> 5), sh = (int)(sbit & 31);
+        uint64_t bits = (uint64_t)hi_in[wi] >> sh;
+        if (sh + exw > 32) bits |= (uint64_t)hi_in[wi + 1] << (32 - sh);
+        bits &= (exw >= 64) ? ~0ull : ((1ull << exw) - 1);
+        out[blk * PFOR_BLK + j] |= (int64_t)(bits << b);
+    }
+}
+
+torch::Tensor pfor_pack(torch::Tensor ints) {
+    CHECK_CUDA(ints);
+    auto v = ints.to(torch::kInt64).contiguous();
+    const int64_t n = v.numel();
+    const int64_t nb = (n + PFOR_BLK - 1) / PFOR_BLK;
+    auto dev = v.device();
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+    auto opts32 = torch::dtype(torch::kInt32).device(dev);
+    auto header = torch::tensor(
+        {(int)(n & 255), (int)((n >> 8) & 255), (int)((n >> 16) & 255),
+         (int)((n >> 24) & 255), 0xFE, (int)(nb & 255), (int)((nb >> 8) & 255),
+         (int)((nb >> 16) & 255), (int)((nb >> 24) & 255)},
+        torch::dtype(torch::kUInt8)).to(dev);
+    if (n == 0) return header;
+
+    auto stats = torch::empty({5 * nb}, opts32);
+    int* width = stats.data_ptr<int>();
+    int* nexc = width + nb;
+    int* excw = nexc + nb;
+    int* sbytes = excw + nb;
+    int* ebytes = sbytes + nb;
+    hipLaunchKernelGGL(pfor_stats_kernel, dim3((int)nb), dim3(WAVE), 0, stream,
+                       v.data_ptr<int64_t>(), n, nb, width, nexc, excw,
+                       sbytes, ebytes);
+    auto sb64 = stats.narrow(0, 3 * nb, nb).to(torch::kInt64);
+    auto eb64 = stats.narrow(0, 4 * nb, nb).to(torch::kInt64);
+    auto sw_off = (sb64.cumsum(0) - sb64) / 4;   // word offsets
+    auto e_off = eb64.cumsum(0) - eb64;
+    const int64_t total_sb = sb64.sum().item<int64_t>();   // host sync
+    const int64_t total_eb = eb64.sum().item<int64_t>();
+    auto sw_off_c = sw_off.contiguous();
+    auto e_off_c = e_off.contiguous();
+
+    // meta bytes [b, ne, exw] per block + pad to 4
+    auto meta = torch::stack({stats.narrow(0, 0, nb), stats.narrow(0, nb, nb),
+                              stats.narrow(0, 2 * nb, nb)}, 1)
+                    .to(torch::kUInt8).reshape({-1});
+    auto headmeta = torch::cat({header, meta});
+    const int64_t hm_pad = (4 - (headmeta.numel() & 3)) & 3;
+    if (hm_pad)
+        headmeta = torch::cat({headmeta,
+                               torch::zeros({hm_pad}, torch::dtype(torch::kUInt8).device(dev))});
+
+    auto body = torch::zeros({total_sb + total_eb},
+                             torch::dtype(torch::kUInt8).device(dev));
+    uint32_t* sw = (uint32_t*)body.data_ptr<uint8_t>();
+    uint8_t* exc = body.data_ptr<uint8_t>() + total_sb;
+    hipLaunchKernelGGL(pfor_pack_kernel, dim3(bt_grid(total_sb / 4)), dim3(256),
+                       0, stream, v.data_ptr<int64_t>(), n, nb, width,
+                       sw_off_c.data_ptr<int64_t>(), total_sb / 4, sw);
+    hipLaunchKernelGGL(pfor_exc_kernel, dim3((int)nb), dim3(WAVE), 0, stream,
+                       v.data_ptr<int64_t>(), n, nb, width, nexc, excw,
+                       e_off_c.data_ptr<int64_t>(), exc);
+    return torch::cat({headmeta, body});
+}
+
+torch::Tensor pfor_unpack(torch::Tensor wire) {
+    CHECK_CUDA(wire);
+    auto w = wire.contiguous();
+    auto head = w.narrow(0, 0, 9).cpu();
+    const uint8_t* h = head.data_ptr<uint8_t>();
+    const int64_t n = (int64_t)h[0] | ((int64_t)h[1] << 8) |
+                      ((int64_t)h[2] << 16) | ((int64_t)h[3] << 24);
+    TORCH_CHECK(h[4] == 0xFE, "not a block-PFoR v2 wire");
+    const int64_t nb = (int64_t)h[5] | ((int64_t)h[6] << 8) |
+                       ((int64_t)h[7] << 16) | ((int64_t)h[8] << 24);
+    auto dev = w.device();
+    auto out = torch::zeros({n}, torch::dtype(torch::kInt64).device(dev));
+    if (n == 0) return out;
+    hipStream_t stream = at::hip::getCurrentHIPStream();
+
+    auto meta = w.narrow(0, 9, nb * 3).to(torch::kInt32).reshape({nb, 3});
+    auto width_t = meta.select(1, 0).contiguous();
+    auto ne_t = meta.select(1, 1).contiguous();
+    auto exw_t = meta.select(1, 2).contiguous();
+    auto cnt = torch::full({nb}, (int64_t)PFOR_BLK,
+                           torch::dtype(torch::kInt64).device(dev));
+    if (n % PFOR_BLK) cnt[nb - 1].fill_(n % PFOR_BLK);
+    auto w64 = width_t.to(torch::kInt64);
+    auto ne64 = ne_t.to(torch::kInt64);
+    auto exw64 = exw_t.to(torch::kInt64);
+    auto sb = ((cnt * w64 + 7).floor_divide(8) + 3).bitwise_and(~3);
+    auto eb = (ne64 + 3).bitwise_and(~3) +
+              ((ne64 * exw64 + 7).floor_divide(8) + 3).bitwise_and(~3);
+    auto sw_off = (sb.cumsum(0) - sb).floor_divide(4).contiguous();
+    auto e_off = (eb.cumsum(0) - eb).contiguous();
+    const int64_t total_sb = sb.sum().item<int64_t>();
+    int64_t hm = 9 + nb * 3;
+    hm += (4 - (hm & 3)) & 3;
+
+    auto body = w.narrow(0, hm, w.numel() - hm).contiguous();
+    const uint32_t* sw = (const uint32_t*)body.data_ptr<uint8_t>();
+    const uint8_t* exc = body.data_ptr<uint8_t>() + total_sb;
+    hipLaunchKernelGGL(pfor_unpack_kernel, dim3(bt_grid(n)), dim3(256), 0,
+                       stream, sw, n, width_t.data_ptr<int>(),
+                       sw_off.data_ptr<int64_t>(), out.data_ptr<int64_t>());
+    hipLaunchKernelGGL(pfor_unexc_kernel, dim3((int)nb), dim3(WAVE), 0, stream,
+                       exc, n, nb, width_t.data_ptr<int>(),
+                       ne_t.data_ptr<int>(), exw_t.data_ptr<int>(),
+                       e_off.data_ptr<int64_t>(), out.data_ptr<int64_t>());
+    return out;
+}
+
 torch::Tensor bloom_insert_cpu(torch::Tensor idxs, int64_t m, int64_t num_hash) {
     auto items = idxs.to(torch::kInt64).contiguous();
     int64_t nbytes = ceil_div(m, 8);
@@ -2944,6 +3217,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "whole-model value-mode compress: polyfit coeffs + int32 idxs");
     m.def("batched_decode_value_sum", &batched_decode_value_sum,
           "multi-rank value-mode decode");
+    m.def("pfor_pack", &pfor_pack, "block-PFoR encode (HIP)");
+    m.def("pfor_unpack", &pfor_unpack, "block-PFoR decode (HIP)");
     m.def("dexp_fit", &dexp_fit,
           "fused DoubleExp cumulative-integral fit (one block per tensor)");
     m.def("bloom_insert_cpu", &bloom_insert_cpu, "Bloom insert (C++ CPU)");

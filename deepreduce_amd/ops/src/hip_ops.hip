@@ -2960,7 +2960,7 @@ torch::Tensor pfor_pack(torch::Tensor ints) {
                        sbytes, ebytes);
     auto sb64 = stats.narrow(0, 3 * nb, nb).to(torch::kInt64);
     auto eb64 = stats.narrow(0, 4 * nb, nb).to(torch::kInt64);
-    auto sw_off = (sb64.cumsum(0) - sb64) / 4;   // word offsets
+    auto sw_off = (sb64.cumsum(0) - sb64).floor_divide(4);  // word offsets
     auto e_off = eb64.cumsum(0) - eb64;
     const int64_t total_sb = sb64.sum().item<int64_t>();   // host sync
     const int64_t total_eb = eb64.sum().item<int64_t>();

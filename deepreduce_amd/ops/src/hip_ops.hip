@@ -2791,7 +2791,9 @@ __global__ void pfor_stats_kernel(const int64_t* __restrict__ v, int64_t n,
     const int lane = threadIdx.x;  // 64 lanes
     const int64_t base = blk * PFOR_BLK;
     const int cnt = (int)min((int64_t)PFOR_BLK, n - base);
-    const int j0 = 2 * lane, j1 = 2 * lane + 1;
+    // lane -> (j, j+64): within each half, ballot lane order == ascending
+    // j, so exception ranks match the torch encoder's flat enumeration
+    const int j0 = lane, j1 = lane + WAVE;
     const int bl0 = (j0 < cnt) ? pfor_bitlen(v[base + j0]) : 0;
     const int bl1 = (j1 < cnt) ? pfor_bitlen(v[base + j1]) : 0;
     // wave max bitlen
@@ -2868,7 +2870,7 @@ __global__ void pfor_exc_kernel(const int64_t* __restrict__ v, int64_t n,
     uint32_t* hi_out = (uint32_t*)(exc + e_off[blk] + ((ne + 3) & ~3));
     int rank_base = 0;
     for (int half = 0; half < 2; ++half) {
-        const int j = 2 * lane + half;
+        const int j = lane + half * WAVE;
         const bool is_exc = (j < cnt) && (pfor_bitlen(v[base + j]) > b);
         const uint64_t ball = __ballot(is_exc);
         if (is_exc) {

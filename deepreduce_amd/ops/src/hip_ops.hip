@@ -1427,17 +1427,21 @@ __global__ void bt_qscatter_own_kernel(const uint64_t* __restrict__ mask,
     }
 }
 
-// decode-side compaction for rank r (launched per rank, sequentially, so the
-// dense accumulation order is deterministic and identical on every rank):
-// dense[voff + i] += wire_r values at the leftmost-k ordinals
-__global__ void bt_qscatter_add_kernel(const uint64_t* __restrict__ mask,
-                                       const int* __restrict__ qoffs, int r,
-                                       int64_t BV, int64_t MW,
-                                       const uint8_t* __restrict__ wires,
-                                       int64_t wstride, int wire_half,
-                                       const int64_t* __restrict__ desc,
-                                       const int* __restrict__ b2t,
-                                       float* __restrict__ dense) {
+// Fused multi-rank decode compaction: ONE candidate-parallel launch
+// replaces R sequential bt_qscatter_add launches.  Each thread owns one
+// universe candidate, walks the R ballot planes, accumulates every
+// rank's wire value for that candidate IN REGISTER, and writes dense[i]
+// once — R read-modify-write passes over the dense buffer become one,
+// and the accumulation order (r ascending within one thread) is
+// deterministic and identical on every rank.
+__global__ void bt_qscatter_add_fused_kernel(const uint64_t* __restrict__ mask,
+                                             const int* __restrict__ qoffs,
+                                             int R, int64_t BV, int64_t MW,
+                                             const uint8_t* __restrict__ wires,
+                                             int64_t wstride, int wire_half,
+                                             const int64_t* __restrict__ desc,
+                                             const int* __restrict__ b2t,
+                                             float* __restrict__ dense) {
     const int t = b2t[blockIdx.x];
     const int64_t* D = bt_row(desc, t);
     const int64_t lb = blockIdx.x - D[10];
@@ -1445,35 +1449,48 @@ __global__ void bt_qscatter_add_kernel(const uint64_t* __restrict__ mask,
     const int64_t end = min(start + BT_CHUNK, D[0]);
     const int64_t k = D[2];
     const int64_t mwoff = D[9];
-    const uint8_t* __restrict__ wbase = wires + r * wstride + D[7];
-    const float* __restrict__ wv = (const float*)wbase;
-    const __half* __restrict__ wh = (const __half*)wbase;
     float* __restrict__ dv = dense + D[1];
-    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
-    __shared__ int wave_cnt[QBLOCK / WAVE];
-    __shared__ int base_s;
-    if (threadIdx.x == 0) base_s = qoffs[r * BV + D[8] + lb];
+    const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    __shared__ int wave_cnt[MAXR][QBLOCK / WAVE];
+    __shared__ int base_s[MAXR];
+    if (threadIdx.x < R)
+        base_s[threadIdx.x] = qoffs[(int64_t)threadIdx.x * BV + D[8] + lb];
     __syncthreads();
     for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
-        int64_t i = i0 + threadIdx.x;
-        uint64_t ball = mask[r * MW + mwoff + ((i0 + (int64_t)wid * WAVE) >> 6)];
-        bool pred = (ball >> lane) & 1;
-        if (lane == 0) wave_cnt[wid] = __popcll(ball);
+        const int64_t i = i0 + threadIdx.x;
+        const int64_t mslot = mwoff + ((i0 + (int64_t)wid * WAVE) >> 6);
+        if (lane == 0)
+            for (int r = 0; r < R; ++r)
+                wave_cnt[r][wid] = __popcll(mask[(int64_t)r * MW + mslot]);
         __syncthreads();
-        if (pred) {
-            int wbase = 0;
-            for (int w = 0; w < wid; ++w) wbase += wave_cnt[w];
-            uint64_t below = (lane == 63) ? (~0ull >> 1) : ((1ull << lane) - 1);
-            int64_t ord = base_s + wbase + __popcll(ball & below);
-            if (ord < k) dv[i] += wire_half ? __half2float(wh[ord]) : wv[ord];
+        float acc = 0.0f;
+        bool any = false;
+        for (int r = 0; r < R; ++r) {
+            const uint64_t ball = mask[(int64_t)r * MW + mslot];
+            if ((ball >> lane) & 1) {
+                int wbase = 0;
+                for (int w = 0; w < wid; ++w) wbase += wave_cnt[r][w];
+                const uint64_t below = (lane == 63) ? (~0ull >> 1)
+                                                    : ((1ull << lane) - 1);
+                const int64_t ord = base_s[r] + wbase + __popcll(ball & below);
+                if (ord < k) {
+                    const uint8_t* wp = wires + (int64_t)r * wstride + D[7];
+                    acc += wire_half
+                               ? __half2float(((const __half*)wp)[ord])
+                               : ((const float*)wp)[ord];
+                    any = true;
+                }
+            }
         }
         __syncthreads();
-        if (threadIdx.x == 0) {
+        if (threadIdx.x < R) {
             int tsum = 0;
-            for (int w = 0; w < QBLOCK / WAVE; ++w) tsum += wave_cnt[w];
-            base_s += tsum;
+            for (int w = 0; w < QBLOCK / WAVE; ++w)
+                tsum += wave_cnt[threadIdx.x][w];
+            base_s[threadIdx.x] += tsum;
         }
         __syncthreads();
+        if (any) dv[i] += acc;
     }
 }
 
@@ -1652,13 +1669,11 @@ torch::Tensor batched_decode_sum(torch::Tensor wires2d, torch::Tensor desc,
     }
     hipLaunchKernelGGL(bt_scan_kernel, dim3(R * T), dim3(QBLOCK), 0, stream,
                        qcounts, d.data_ptr<int64_t>(), T, BV, qoffs);
-    for (int r = 0; r < R; ++r) {
-        hipLaunchKernelGGL(bt_qscatter_add_kernel, dim3((int)BV), dim3(QBLOCK), 0, stream,
-                           (const uint64_t*)mask.data_ptr<int64_t>(), qoffs, r, BV,
-                           mask_words, w.data_ptr<uint8_t>(), W, (int)wire_half,
-                           d.data_ptr<int64_t>(), map.data_ptr<int>(),
-                           dense.data_ptr<float>());
-    }
+    hipLaunchKernelGGL(bt_qscatter_add_fused_kernel, dim3((int)BV), dim3(QBLOCK),
+                       0, stream, (const uint64_t*)mask.data_ptr<int64_t>(),
+                       qoffs, R, BV, mask_words, w.data_ptr<uint8_t>(), W,
+                       (int)wire_half, d.data_ptr<int64_t>(),
+                       map.data_ptr<int>(), dense.data_ptr<float>());
     return dense;
 }
 

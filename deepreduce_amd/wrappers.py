@@ -281,6 +281,11 @@ class DeepReduce(_WrapperBase):
         count without it (p0 carries it in-band; topk/randomk imply k)."""
         if not getattr(self.val_codec, "order_preserving", False):
             return False
+        # the FP-aware re-read puts values in ascending-position order —
+        # exactly the order decompress re-derives; without it the values
+        # stay in topk-magnitude order and the mapping must travel
+        if not self.params.get("fp_aware", True):
+            return False
         return (self.params.get("policy", "leftmost") == "p0"
                 or getattr(self.sparsifier, "compress_ratio", None) is not None)
 

@@ -87,7 +87,7 @@ def main():
     b = bp.decode_sum(stacked)
     results["decode_parity"] = bool(torch.equal(a, b))
     results["ldsq_bytes"] = saved
-    results["qhead"] = os.environ.get("DEEPREDUCE_QHEAD", "4")
+    results["qhead"] = os.environ.get("DEEPREDUCE_QHEAD", "2")
     print(json.dumps(results))
 
 

@@ -360,3 +360,29 @@ def test_auto_fpr():
     validate({"fpr": "auto"})
     with pytest.raises(ValueError):
         validate({"fpr": "wrong"})
+
+
+def test_generic_decompress_batch_chunking_cpu():
+    """IndexCompressor.decompress_batch chunks >16 payloads (the MAXR
+    kernel cap) on the CPU native path too."""
+    from deepreduce_amd import deepreduce_from_params
+
+    grc = deepreduce_from_params({
+        "compressor": "topk", "memory": "none",
+        "communicator": "allgather", "compress_ratio": 0.02,
+        "deepreduce": "index", "index": "bloom", "policy": "leftmost",
+    })
+    comp = grc.compressor
+    torch.manual_seed(41)
+    N = 20_000
+    payloads, ref = [], torch.zeros(N)
+    ctx = None
+    for r in range(18):
+        t = torch.randn(N)
+        tc, ctx = comp.compress(t, f"r{r}")
+        payloads.append(tc)
+        ref += comp.decompress(tc, ctx).reshape(-1)
+    got = comp.decompress_batch(payloads, ctx)
+    assert got is not None, "fast path must engage for 18 ranks"
+    assert torch.allclose(got.reshape(-1), ref, atol=1e-5), \
+        (got.reshape(-1) - ref).abs().max()

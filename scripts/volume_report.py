@@ -29,12 +29,18 @@ CONFIGS = [
     ("DR-BF (leftmost)", {"deepreduce": "index", "index": "bloom",
                           "policy": "leftmost"}),
     ("DR-BF-P0", {"deepreduce": "index", "index": "bloom", "policy": "p0"}),
+    ("DR-BF-P0 fpr=auto", {"deepreduce": "index", "index": "bloom",
+                           "policy": "p0", "fpr": "auto"}),
     ("DR-FitPoly", {"deepreduce": "value", "value": "polyfit"}),
     ("DR-QSGD", {"deepreduce": "value", "value": "qsgd"}),
     ("DR-RLE", {"deepreduce": "index", "index": "rle"}),
     ("DR-Huffman", {"deepreduce": "index", "index": "huffman"}),
     ("DR-Gzip", {"deepreduce": "value", "value": "gzip"}),
     ("DR-PFor", {"deepreduce": "index", "index": "pfor"}),
+    ("DR-QSGD+BF-P0 7bit auto", {"deepreduce": "both", "value": "qsgd",
+                                 "index": "bloom", "policy": "p0",
+                                 "fpr": "auto", "qsgd_pack": True,
+                                 "quantum_num": 63}),
     ("DR-QSGD+BF-P0 ('both')", {"deepreduce": "both", "value": "qsgd",
                                 "index": "bloom", "policy": "p0"}),
     ("DR-FitPoly+BF ('both')", {"deepreduce": "both", "value": "polyfit",

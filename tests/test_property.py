@@ -142,3 +142,38 @@ def test_wrapper_step_invariants(mode, data):
     r = grc.memory.residuals["w"]
     assert torch.allclose(r, t - out.view_as(t), atol=2e-4), \
         (r - (t - out.view_as(t))).abs().max()
+
+
+@settings(max_examples=30, deadline=None)
+@given(data=st.data())
+def test_federated_payload_serialization_fuzz(data):
+    """serialize/deserialize round-trips arbitrary payload dicts (dtype
+    mix, zero-length chunks, many tensors)."""
+    from deepreduce_amd.federated_dist import (deserialize_payloads,
+                                               serialize_payloads)
+
+    dtypes = [torch.float32, torch.float64, torch.float16, torch.int64,
+              torch.int32, torch.int8, torch.uint8]
+    n_names = data.draw(st.integers(1, 6))
+    names = [f"p{i}" for i in range(n_names)]
+    g = torch.Generator().manual_seed(data.draw(st.integers(0, 2**31 - 1)))
+    payloads = {}
+    for n in names:
+        n_chunks = data.draw(st.integers(1, 4))
+        chunks = []
+        for _ in range(n_chunks):
+            dt = dtypes[data.draw(st.integers(0, len(dtypes) - 1))]
+            numel = data.draw(st.integers(0, 300))
+            if dt.is_floating_point:
+                t = torch.randn(numel, generator=g).to(dt)
+            else:
+                t = torch.randint(0, 100, (numel,), generator=g).to(dt)
+            chunks.append(t)
+        payloads[n] = tuple(chunks)
+    buf = serialize_payloads(payloads, names)
+    out = deserialize_payloads(buf, names)
+    for n in names:
+        assert len(out[n]) == len(payloads[n])
+        for x, y in zip(out[n], payloads[n]):
+            assert x.dtype == y.dtype
+            assert torch.equal(x, y.reshape(-1))
